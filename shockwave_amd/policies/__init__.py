@@ -1,0 +1,60 @@
+"""Policy registry (reference: utils.get_policy, utils.py:603-686)."""
+
+from .base import Policy, PolicyWithPacking
+from .simple import IsolatedPolicy, ProportionalPolicy, GandivaFairPolicy
+from .fifo import FIFOPolicy, FIFOPolicyWithPerf, FIFOPolicyWithPacking
+from .max_min_fairness import (
+    MaxMinFairnessPolicy,
+    MaxMinFairnessPolicyWithPerf,
+    MaxMinFairnessWaterFillingPolicy,
+    MaxMinFairnessStrategyProofPolicy,
+)
+from .finish_time_fairness import (
+    FinishTimeFairnessPolicy,
+    FinishTimeFairnessPolicyWithPerf,
+)
+from .max_sum_throughput import (
+    ThroughputSumWithPerf,
+    ThroughputNormalizedByCostSumWithPerf,
+    ThroughputNormalizedByCostSumWithPerfSLOs,
+)
+from .min_total_duration import MinTotalDurationPolicy, MinTotalDurationPolicyWithPerf
+from .allox import AlloXPolicy
+from .gandiva import GandivaPolicy
+
+
+class ShockwavePolicyStub(Policy):
+    """Name marker only: the Shockwave planner drives the round schedule
+    directly (engine + solver/planner.py), matching the reference where
+    policies/shockwave.py is a 10-line stub."""
+
+    name = "Shockwave"
+
+
+def get_policy(policy_name: str, seed=None, solver=None,
+               priority_reweighting_policies=None):
+    table = {
+        "allox": lambda: AlloXPolicy(),
+        "fifo": lambda: FIFOPolicy(seed=seed),
+        "fifo_perf": FIFOPolicyWithPerf,
+        "fifo_packed": FIFOPolicyWithPacking,
+        "finish_time_fairness": FinishTimeFairnessPolicy,
+        "finish_time_fairness_perf": FinishTimeFairnessPolicyWithPerf,
+        "gandiva": lambda: GandivaPolicy(seed=seed),
+        "gandiva_fair": GandivaFairPolicy,
+        "isolated": IsolatedPolicy,
+        "isolated_plus": IsolatedPolicy,
+        "max_min_fairness": MaxMinFairnessPolicy,
+        "max_min_fairness_perf": MaxMinFairnessPolicyWithPerf,
+        "max_min_fairness_water_filling": MaxMinFairnessWaterFillingPolicy,
+        "max_min_fairness_strategy_proof": MaxMinFairnessStrategyProofPolicy,
+        "max_sum_throughput_perf": ThroughputSumWithPerf,
+        "max_sum_throughput_normalized_by_cost_perf": ThroughputNormalizedByCostSumWithPerf,
+        "max_sum_throughput_normalized_by_cost_perf_SLOs": ThroughputNormalizedByCostSumWithPerfSLOs,
+        "min_total_duration": MinTotalDurationPolicy,
+        "min_total_duration_perf": MinTotalDurationPolicyWithPerf,
+        "shockwave": ShockwavePolicyStub,
+    }
+    if policy_name not in table:
+        raise ValueError(f"unknown policy {policy_name!r}")
+    return table[policy_name]()

@@ -1,0 +1,197 @@
+"""Max-min fairness (Gavel's LAS) policies via LP.
+
+Reference: policies/max_min_fairness.py:1-410.  Maximize the minimum
+(priority- and share-normalized) effective throughput across jobs:
+
+    max min_i sum_j (w_ij * x_ij)     w_ij = tput_ij * prio_i * sf_i
+
+LAS variant replaces throughputs with 1.0 so allocations equalize *time*
+share rather than throughput (max_min_fairness.py:33-44).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from .base import Policy
+from .simple import ProportionalPolicy
+
+
+class MaxMinFairnessPolicyWithPerf(Policy):
+    name = "MaxMinFairness_Perf"
+
+    def __init__(self):
+        self._proportional = ProportionalPolicy()
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+    ):
+        throughputs, index = self.flatten(unflattened_throughputs, cluster_spec)
+        if throughputs is None:
+            return None
+        m, n = throughputs.shape
+        job_ids, worker_types = index
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+
+        priority = np.array(
+            [1.0 / unflattened_priority_weights[jid] for jid in job_ids]
+        )
+        proportional_tputs = self._proportional.get_throughputs(
+            throughputs, index, cluster_spec
+        )
+        priority = priority.reshape((m, 1)) / proportional_tputs.reshape((m, 1))
+
+        weights = throughputs * priority * sfa
+
+        # LP vars: [x flattened (m*n), t]; maximize t
+        nv = m * n + 1
+        A_ub, b_ub = self.base_constraints(m, n, sfa, extra_vars=1)
+        rows = []
+        for i in range(m):
+            row = np.zeros(nv)
+            row[i * n : (i + 1) * n] = -weights[i]
+            row[-1] = 1.0  # t - w_i . x_i <= 0
+            rows.append(row)
+        A_ub = np.vstack([A_ub, np.array(rows)])
+        b_ub = np.concatenate([b_ub, np.zeros(m)])
+        c = np.zeros(nv)
+        c[-1] = -1.0
+        res = self.solve_lp(c, A_ub, b_ub)
+        if not res.success:
+            return None
+        x = self.clip_allocation(res.x[: m * n].reshape((m, n)))
+        return self.unflatten(x, index)
+
+
+class MaxMinFairnessPolicy(Policy):
+    """LAS: unit throughputs (time-fair, not throughput-fair)."""
+
+    name = "MaxMinFairness"
+
+    def __init__(self):
+        self._perf = MaxMinFairnessPolicyWithPerf()
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        priority_weights,
+        cluster_spec,
+    ):
+        ones = {
+            jid: {wt: 1.0 for wt in per_wt}
+            for jid, per_wt in unflattened_throughputs.items()
+        }
+        return self._perf.get_allocation(
+            ones, scale_factors, priority_weights, cluster_spec
+        )
+
+
+class MaxMinFairnessWaterFillingPolicy(Policy):
+    """Water-filling max-min fairness: iteratively maximize the minimum
+    normalized rate, freeze saturated jobs at the achieved level, recurse on
+    the rest (reference max_min_fairness_water_filling.py, 718 lines; this
+    is the standard algorithm, not a translation)."""
+
+    name = "MaxMinFairnessWaterFilling"
+
+    def __init__(self, priority_reweighting_policies=None, max_iterations=64):
+        self._max_iterations = max_iterations
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+        **kwargs,
+    ):
+        throughputs, index = self.flatten(unflattened_throughputs, cluster_spec)
+        if throughputs is None:
+            return None
+        m, n = throughputs.shape
+        job_ids, worker_types = index
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+        priority = np.array(
+            [1.0 / unflattened_priority_weights[jid] for jid in job_ids]
+        ).reshape((m, 1))
+        weights = throughputs * priority * sfa
+
+        frozen_level = {}  # i -> minimum rate locked in
+        x_final = np.zeros((m, n))
+        for _ in range(self._max_iterations):
+            free = [i for i in range(m) if i not in frozen_level]
+            if not free:
+                break
+            nv = m * n + 1
+            A_ub, b_ub = self.base_constraints(m, n, sfa, extra_vars=1)
+            rows, rhs = [], []
+            for i in free:
+                row = np.zeros(nv)
+                row[i * n : (i + 1) * n] = -weights[i]
+                row[-1] = 1.0
+                rows.append(row)
+                rhs.append(0.0)
+            for i, level in frozen_level.items():
+                row = np.zeros(nv)
+                row[i * n : (i + 1) * n] = -weights[i]
+                rows.append(row)
+                rhs.append(-level)
+            A_ub = np.vstack([A_ub, np.array(rows)])
+            b_ub = np.concatenate([b_ub, np.array(rhs)])
+            c = np.zeros(nv)
+            c[-1] = -1.0
+            res = self.solve_lp(c, A_ub, b_ub)
+            if not res.success:
+                break
+            t_star = res.x[-1]
+            x_star = res.x[: m * n].reshape((m, n))
+            x_final = x_star
+            # freeze jobs that can't exceed t_star: test each free job by
+            # checking if its rate can be raised with others fixed >= t_star
+            newly_frozen = False
+            for i in free:
+                c2 = np.zeros(nv)
+                c2[i * n : (i + 1) * n] = -weights[i]
+                rows2 = []
+                rhs2 = []
+                for k in free:
+                    if k == i:
+                        continue
+                    row = np.zeros(nv)
+                    row[k * n : (k + 1) * n] = -weights[k]
+                    rows2.append(row)
+                    rhs2.append(-t_star + 1e-9)
+                for k, level in frozen_level.items():
+                    row = np.zeros(nv)
+                    row[k * n : (k + 1) * n] = -weights[k]
+                    rows2.append(row)
+                    rhs2.append(-level)
+                A2 = np.vstack([A_ub[: n + m], np.array(rows2)]) if rows2 else A_ub[: n + m]
+                b2 = (
+                    np.concatenate([b_ub[: n + m], np.array(rhs2)])
+                    if rows2
+                    else b_ub[: n + m]
+                )
+                res2 = self.solve_lp(c2, A2, b2)
+                best = -res2.fun if res2.success else t_star
+                if best <= t_star * (1 + 1e-6) + 1e-9:
+                    frozen_level[i] = t_star
+                    newly_frozen = True
+            if not newly_frozen:
+                for i in free:
+                    frozen_level[i] = t_star
+        return self.unflatten(self.clip_allocation(x_final), index)
+
+
+class MaxMinFairnessStrategyProofPolicy(MaxMinFairnessPolicy):
+    """Strategy-proof variant: LAS allocation computed on reported
+    throughputs but normalized shares use unit demands, which removes the
+    incentive to misreport (reference max_min_fairness_strategy_proof.py).
+    With a single homogeneous worker type this coincides with LAS."""
+
+    name = "MaxMinFairness_StrategyProof"

@@ -1,0 +1,226 @@
+"""Policy allocation tests: LP policies checked against closed-form /
+brute-force expectations on small instances."""
+
+import numpy as np
+import pytest
+
+from shockwave_amd.core.job import JobIdPair
+from shockwave_amd.policies import (
+    AlloXPolicy,
+    FIFOPolicy,
+    FinishTimeFairnessPolicy,
+    GandivaFairPolicy,
+    GandivaPolicy,
+    IsolatedPolicy,
+    MaxMinFairnessPolicy,
+    MaxMinFairnessPolicyWithPerf,
+    MaxMinFairnessWaterFillingPolicy,
+    MinTotalDurationPolicy,
+    ProportionalPolicy,
+    ThroughputSumWithPerf,
+    get_policy,
+)
+
+WT = "mi355x"
+
+
+def mk_tputs(vals):
+    return {JobIdPair(i): {WT: v} for i, v in enumerate(vals)}
+
+
+def mk_sf(n, sf=1):
+    return {JobIdPair(i): sf for i in range(n)}
+
+
+def mk_prio(n):
+    return {JobIdPair(i): 1.0 for i in range(n)}
+
+
+def total_workers_used(alloc, sf):
+    return sum(alloc[j][WT] * sf[j] for j in alloc)
+
+
+class TestIsolatedProportional:
+    def test_isolated_equal_split(self):
+        alloc = IsolatedPolicy().get_allocation(mk_tputs([1, 1]), mk_sf(2), {WT: 4})
+        # 2 jobs, 4 GPUs -> each gets min(1, 2) = 1.0
+        for j in alloc:
+            assert alloc[j][WT] == pytest.approx(1.0)
+
+    def test_isolated_oversubscribed(self):
+        alloc = IsolatedPolicy().get_allocation(mk_tputs([1] * 8), mk_sf(8), {WT: 4})
+        for j in alloc:
+            assert alloc[j][WT] == pytest.approx(0.5)
+
+    def test_proportional(self):
+        # reference semantics (proportional.py:38-41): rows normalized by the
+        # max row sum, so each row sums to exactly 1 with a single worker type
+        alloc = ProportionalPolicy().get_allocation(mk_tputs([1] * 4), {WT: 2})
+        for j in alloc:
+            assert alloc[j][WT] == pytest.approx(1.0)
+
+    def test_gandiva_fair(self):
+        alloc = GandivaFairPolicy().get_allocation(
+            mk_tputs([1] * 4), mk_sf(4), {WT: 2}
+        )
+        for j in alloc:
+            assert alloc[j][WT] == pytest.approx(0.5)
+
+
+class TestMaxMinFairness:
+    def test_las_equalizes_time(self):
+        # LAS ignores throughputs: equal time share
+        alloc = MaxMinFairnessPolicy().get_allocation(
+            mk_tputs([10.0, 1.0]), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(0.5, abs=1e-4)
+        assert alloc[JobIdPair(1)][WT] == pytest.approx(0.5, abs=1e-4)
+
+    def test_perf_equalizes_throughput_ratio(self):
+        # with perf, allocations weight toward the slow job to equalize
+        # normalized effective throughput
+        alloc = MaxMinFairnessPolicyWithPerf().get_allocation(
+            mk_tputs([4.0, 1.0]), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        x0, x1 = alloc[JobIdPair(0)][WT], alloc[JobIdPair(1)][WT]
+        assert x0 + x1 <= 1.0 + 1e-6
+        # normalized rates equal: (tput0*x0)/prop0 == (tput1*x1)/prop1
+        # prop_i = tput_i * 0.5 -> x0 == x1 == 0.5
+        assert x0 == pytest.approx(0.5, abs=1e-3)
+
+    def test_scale_factor_capacity(self):
+        sf = {JobIdPair(0): 4, JobIdPair(1): 1}
+        alloc = MaxMinFairnessPolicy().get_allocation(
+            mk_tputs([1.0, 1.0]), sf, mk_prio(2), {WT: 4}
+        )
+        used = alloc[JobIdPair(0)][WT] * 4 + alloc[JobIdPair(1)][WT] * 1
+        assert used <= 4 + 1e-6
+
+    def test_water_filling_matches_maxmin_on_symmetric(self):
+        alloc = MaxMinFairnessWaterFillingPolicy().get_allocation(
+            mk_tputs([1.0, 1.0]), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(0.5, abs=1e-3)
+
+    def test_water_filling_uses_slack(self):
+        # 3 jobs, 2 GPUs: max-min level is 2/3, but water-filling should
+        # not leave capacity stranded
+        alloc = MaxMinFairnessWaterFillingPolicy().get_allocation(
+            mk_tputs([1.0, 1.0, 1.0]), mk_sf(3), mk_prio(3), {WT: 2}
+        )
+        used = sum(alloc[j][WT] for j in alloc)
+        assert used == pytest.approx(2.0, abs=1e-2)
+
+
+class TestThemis:
+    def test_equal_jobs_equal_rho(self):
+        p = FinishTimeFairnessPolicy()
+        n = 2
+        alloc = p.get_allocation(
+            mk_tputs([2.0, 2.0]),
+            mk_sf(n),
+            mk_prio(n),
+            {JobIdPair(i): 100.0 for i in range(n)},
+            {JobIdPair(i): 1000.0 for i in range(n)},
+            {WT: 1},
+        )
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(0.5, abs=0.05)
+
+    def test_lagging_job_favored(self):
+        p = FinishTimeFairnessPolicy()
+        times = {JobIdPair(0): 5000.0, JobIdPair(1): 100.0}
+        steps = {JobIdPair(0): 1000.0, JobIdPair(1): 1000.0}
+        alloc = p.get_allocation(
+            mk_tputs([2.0, 2.0]), mk_sf(2), mk_prio(2), times, steps, {WT: 1}
+        )
+        # job 0 has waited much longer -> needs more share to hit same rho
+        assert alloc[JobIdPair(0)][WT] > alloc[JobIdPair(1)][WT]
+
+
+class TestMST:
+    def test_prefers_fast_job(self):
+        alloc = ThroughputSumWithPerf().get_allocation(
+            mk_tputs([10.0, 1.0]), mk_sf(2), {WT: 1}
+        )
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(1.0, abs=1e-4)
+
+    def test_capacity_respected(self):
+        alloc = ThroughputSumWithPerf().get_allocation(
+            mk_tputs([5.0, 4.0, 3.0]), mk_sf(3), {WT: 2}
+        )
+        assert total_workers_used(alloc, mk_sf(3)) <= 2 + 1e-6
+
+
+class TestOSSP:
+    def test_binary_search_feasible(self):
+        steps = {JobIdPair(0): 1000.0, JobIdPair(1): 4000.0}
+        alloc = MinTotalDurationPolicy().get_allocation(
+            mk_tputs([2.0, 2.0]), mk_sf(2), steps, {WT: 1}
+        )
+        # both jobs must fit in the same horizon T*:
+        # T* = (1000+4000)/2 = 2500s; job1 needs 4x job0's share
+        x0, x1 = alloc[JobIdPair(0)][WT], alloc[JobIdPair(1)][WT]
+        assert x1 / max(x0, 1e-9) == pytest.approx(4.0, rel=0.15)
+
+
+class TestFIFO:
+    def test_arrival_order(self):
+        p = FIFOPolicy(seed=0)
+        alloc = p.get_allocation(mk_tputs([1, 1, 1]), mk_sf(3), {WT: 2})
+        assert alloc[JobIdPair(0)][WT] == 1.0
+        assert alloc[JobIdPair(1)][WT] == 1.0
+        assert alloc[JobIdPair(2)][WT] == 0.0
+
+    def test_sticky_until_done(self):
+        p = FIFOPolicy(seed=0)
+        p.get_allocation(mk_tputs([1, 1, 1]), mk_sf(3), {WT: 2})
+        # job 0 completes; job 2 takes its slot
+        tputs = {JobIdPair(1): {WT: 1}, JobIdPair(2): {WT: 1}}
+        sf = {JobIdPair(1): 1, JobIdPair(2): 1}
+        alloc = p.get_allocation(tputs, sf, {WT: 2})
+        assert alloc[JobIdPair(1)][WT] == 1.0
+        assert alloc[JobIdPair(2)][WT] == 1.0
+
+
+class TestAllox:
+    def test_assigns_workers(self):
+        p = AlloXPolicy()
+        n = 3
+        alloc = p.get_allocation(
+            mk_tputs([2.0, 1.0, 1.5]),
+            mk_sf(n),
+            {JobIdPair(i): float(i) for i in range(n)},
+            {JobIdPair(i): 100.0 for i in range(n)},
+            [],
+            {WT: 2},
+        )
+        placed = [j for j in alloc if alloc[j][WT] > 0]
+        assert len(placed) == 2
+
+
+class TestGandiva:
+    def test_undersubscribed_proportional(self):
+        p = GandivaPolicy(seed=0)
+        tputs = {JobIdPair(0): {WT: 1.0}, JobIdPair(1): {WT: 1.0}}
+        sf = {JobIdPair(0): 1, JobIdPair(1): 1}
+        alloc = p.get_allocation(tputs, sf, {WT: 4})
+        assert alloc[JobIdPair(0)][WT] > 0
+
+
+class TestFactory:
+    @pytest.mark.parametrize(
+        "name",
+        [
+            "fifo", "fifo_perf", "fifo_packed", "finish_time_fairness",
+            "gandiva", "gandiva_fair", "isolated", "max_min_fairness",
+            "max_min_fairness_perf", "max_min_fairness_water_filling",
+            "max_sum_throughput_perf", "min_total_duration", "shockwave",
+            "allox",
+        ],
+    )
+    def test_factory(self, name):
+        assert get_policy(name) is not None
+
+    def test_unknown_raises(self):
+        with pytest.raises(ValueError):
+            get_policy("nope")
