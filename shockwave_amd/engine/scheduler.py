@@ -1,0 +1,1357 @@
+"""Round-based cluster scheduler: simulation + physical mechanism core.
+
+Rebuild of the reference's Scheduler (scheduler/scheduler.py, 4931 lines)
+around the same observable behavior:
+
+* event-driven **simulation** replaying a trace against oracle throughputs
+  (reference ``simulate()`` :1728-2268),
+* greedy priority-sorted **round scheduling** for Gavel policies and
+  planner-driven scheduling for Shockwave
+  (``_schedule_jobs_on_workers_helper`` :1113-1273),
+* per-round micro-task accounting with EMA throughput updates, failure
+  tracking, deadline aborts (``_done_callback`` :4341-4729),
+* Accordion/GNS **simulation twins** flipping batch-size flags from the
+  oracle schedules (``_simulate_accordion``/``_simulate_gns`` :1604-1726)
+  and epoch-preserving rescaling (``_scale_bs_and_iters`` :4731-4931),
+* the metric suite: JCT (avg/geo/harmonic), makespan, finish-time fairness
+  (static + Themis contention), envy, utilization, lease-extension rate.
+
+The physical mode (gRPC workers, leases, dispatch) lives in
+``shockwave_amd/engine/physical.py`` and shares this state machine.
+"""
+
+from __future__ import annotations
+
+import collections
+import copy
+import heapq
+import logging
+import math
+import random
+import time
+from collections import OrderedDict
+from typing import Dict, List, Optional
+
+import numpy as np
+import scipy.stats
+
+from ..core import datasets
+from ..core.job import Job, JobIdPair
+from ..core.metadata import JobMetadata
+from ..solver.planner import ShockwavePlanner
+
+# Constants (reference scheduler.py:41-71)
+INFINITY = int(1e9)
+DEFAULT_THROUGHPUT = 1
+EMA_ALPHA = 0.5
+MAX_FAILED_ATTEMPTS = 5
+SCHEDULE_RECOMPUTE_FRACTION = 0.5
+JOB_COMPLETION_BUFFER_TIME = 60
+BASE_JOB_PORT = 60570
+MAX_PORT = 65535
+EARLY_INIT_THRESHOLD = 3.0
+REOPT_ROUNDS = 8
+PREEMPTION_OVERHEAD_S = 20  # NFS checkpoint/restore cost injected on migration
+
+logger = logging.getLogger("shockwave_amd.engine")
+
+
+class RoundScheduler:
+    def __init__(
+        self,
+        policy,
+        simulate: bool = True,
+        throughputs: Optional[Dict] = None,
+        seed: int = 0,
+        time_per_iteration: float = 120.0,
+        minimum_time_between_allocation_resets: float = 1000.0,
+        max_rounds: Optional[int] = None,
+        profiles: Optional[List[Dict]] = None,
+        shockwave_config: Optional[Dict] = None,
+        worker_type: str = "mi355x",
+    ):
+        self._policy = policy
+        self._simulate = simulate
+        self._oracle_throughputs = throughputs
+        self._time_per_iteration = time_per_iteration
+        self._minimum_time_between_allocation_resets = (
+            minimum_time_between_allocation_resets
+        )
+        self._max_rounds = max_rounds
+        self._profiles = profiles
+        self._worker_type = worker_type
+        self._job_packing = "Packing" in getattr(policy, "name", "")
+
+        self._start_timestamp = 0.0 if simulate else time.time()
+        self._current_timestamp = self._start_timestamp
+
+        self._init_seeds(seed)
+
+        # job state
+        self._job_id_counter = 0
+        self._jobs: Dict[JobIdPair, Job] = OrderedDict()
+        self._throughputs: Dict = {}
+        self._steps_run_so_far: Dict = {}
+        self._total_steps_run: Dict = {}
+        self._job_time_so_far: Dict = {}
+        self._job_cost_so_far: Dict = {}
+        self._cumulative_run_time: Dict = {}
+        self._num_failures_per_job: Dict = {}
+        self._per_job_start_timestamps: Dict = {}
+        self._per_job_latest_timestamps: Dict = {}
+        self._job_completion_times: Dict = {}
+        self._job_priority_weights: Dict = {}
+        self._completed_jobs = set()
+        self._running_jobs = set()
+        self._steps_run_in_current_lease: Dict = {}
+        self._original_bs: Dict = {}
+        self._original_num_steps: Dict = {}
+        self._job_types: Dict = {}
+        self._bs_flags: Dict = {}
+        self._num_jobs_in_trace = 0
+        self._in_progress_updates: Dict = collections.defaultdict(list)
+        self._lease_update_requests: Dict = collections.defaultdict(list)
+        self._max_steps: Dict = {}
+        self._jobs_with_extended_lease = set()
+        self._job_timelines: Dict = {}
+
+        # worker state
+        self._worker_ids: List[int] = []
+        self._worker_types: List[str] = []
+        self._cluster_spec: Dict[str, int] = {}
+        self._worker_id_to_worker_type_mapping: Dict[int, str] = {}
+        self._worker_type_to_worker_id_mapping: Dict[str, List[List[int]]] = {}
+        self._worker_start_times: Dict = {}
+        self._cumulative_worker_time_so_far: Dict = {}
+        self._worker_time_so_far: Dict = {}
+
+        # scheduling state
+        self._allocation: Dict = {}
+        self._priorities: Dict = {}
+        self._deficits: Dict = {}
+        self._need_to_update_allocation = True
+        self._allocation_changed_since_last_time_reset = False
+        self._last_reset_time = 0.0
+        self._current_worker_assignments: "OrderedDict" = OrderedDict()
+        self._num_completed_rounds = 0
+        self._num_lease_extensions = 0
+        self._num_lease_extension_opportunities = 0
+        self._per_round_schedule: List[Dict] = []
+        self._num_jobs_in_curr_round: List[int] = []
+        self._num_scheduled_rounds: Dict = {}
+        self._num_queued_rounds: Dict = {}
+        self._job_start_round: Dict = {}
+        self._job_end_round: Dict = {}
+        self._throughput_timeline: Dict = {}
+
+        # shockwave state
+        self._shockwave_planner: Optional[ShockwavePlanner] = None
+        self._scheduled_jobs_in_current_round: List = []
+        self._scheduled_jobs_in_prev_round: List = []
+        self._iround_reopt = 0
+        self._shockwave_job_completed_flag = False
+        if self.is_shockwave:
+            assert shockwave_config is not None, "shockwave needs a config"
+            assert profiles is not None, "shockwave needs job profiles"
+            self._shockwave_config = shockwave_config
+            self._shockwave_planner = ShockwavePlanner(
+                ngpus=shockwave_config["num_gpus"],
+                gram=shockwave_config.get("gpu_ram", 288),
+                init_metadata=OrderedDict(),
+                future_nrounds=shockwave_config.get("future_rounds", 20),
+                round_duration=shockwave_config.get(
+                    "time_per_iteration", time_per_iteration
+                ),
+                solver_rel_gap=shockwave_config.get("solver_rel_gap", 1e-3),
+                solver_timeout=shockwave_config.get("solver_timeout", 15),
+                logapx_bases=shockwave_config.get(
+                    "log_approximation_bases", [0.0, 0.2, 0.4, 0.6, 0.8, 1.0]
+                ),
+                logapx_origin={0.0: shockwave_config.get("logapx_origin", 1e-1)},
+                k=shockwave_config.get("k", 1e-3),
+                lam=shockwave_config.get("lambda", 12.0),
+                rhomax=shockwave_config.get("rhomax", 1.0),
+            )
+
+    # ------------------------------------------------------------------
+    @property
+    def is_shockwave(self) -> bool:
+        return getattr(self._policy, "name", "").lower() == "shockwave"
+
+    def _init_seeds(self, seed):
+        self._job_generator = random.Random(seed)
+        self._interarrival_time_generator = random.Random(seed + 1)
+        self._worker_type_shuffler = random.Random(seed + 2)
+        self._SLO_generator = random.Random(seed + 3)
+
+    def get_current_timestamp(self, in_seconds=False):
+        if self._simulate:
+            return self._current_timestamp
+        if in_seconds:
+            return time.time() - self._start_timestamp
+        return time.time()
+
+    # ------------------------------------------------------------------
+    # Worker registration (simulation shortcut of the RPC callback,
+    # reference _register_worker_callback :3782-3878)
+    # ------------------------------------------------------------------
+
+    def register_worker(self, worker_type: str, num_gpus: int = 1):
+        if worker_type not in self._worker_type_to_worker_id_mapping:
+            self._worker_types.append(worker_type)
+            self._worker_types.sort()
+            self._worker_type_to_worker_id_mapping[worker_type] = []
+            self._cluster_spec[worker_type] = 0
+            self._worker_time_so_far[worker_type] = 0.0
+            self._priorities[worker_type] = {}
+            self._deficits[worker_type] = {}
+            # late-registered worker types need throughput entries
+            for job_id in self._jobs:
+                self._steps_run_so_far[job_id].setdefault(worker_type, 0)
+                self._set_initial_throughput(job_id, worker_type)
+        server_worker_ids = []
+        for _ in range(num_gpus):
+            worker_id = len(self._worker_ids)
+            self._worker_ids.append(worker_id)
+            self._worker_id_to_worker_type_mapping[worker_id] = worker_type
+            self._cluster_spec[worker_type] += 1
+            self._worker_start_times[worker_id] = self.get_current_timestamp()
+            self._cumulative_worker_time_so_far[worker_id] = 0.0
+            server_worker_ids.append(worker_id)
+        self._worker_type_to_worker_id_mapping[worker_type].append(
+            server_worker_ids
+        )
+        self._need_to_update_allocation = True
+        return server_worker_ids
+
+    # ------------------------------------------------------------------
+    # Job lifecycle
+    # ------------------------------------------------------------------
+
+    def _set_initial_throughput(self, job_id: JobIdPair, worker_type: str):
+        job = self._jobs[job_id]
+        key = (job.job_type, job.scale_factor)
+        if self._oracle_throughputs is not None:
+            self._throughputs[job_id][worker_type] = self._oracle_throughputs[
+                worker_type
+            ][key]["null"]
+        else:
+            self._throughputs[job_id][worker_type] = DEFAULT_THROUGHPUT
+
+    def _populate_job_combination_metadata(self, job_id, worker_type):
+        """Pair throughputs for packing policies (reference :988-1046)."""
+        job = self._jobs[job_id]
+        for other_job_id in list(self._jobs.keys()):
+            if other_job_id == job_id:
+                continue
+            other_job = self._jobs[other_job_id]
+            if job.scale_factor != other_job.scale_factor:
+                continue
+            merged = JobIdPair(job_id[0], other_job_id[0])
+            if merged not in self._throughputs:
+                self._throughputs[merged] = {}
+                self._job_time_so_far[merged] = {
+                    wt: self._time_per_iteration / 2.0
+                    for wt in self._worker_types
+                }
+            key = (job.job_type, job.scale_factor)
+            other_key = (other_job.job_type, other_job.scale_factor)
+            oracle = self._oracle_throughputs[worker_type]
+            if key in oracle and other_key in oracle.get(key, {}):
+                pair = oracle[key][other_key]
+            else:
+                pair = [0.0, 0.0]
+            if merged.singletons()[0] == job_id:
+                self._throughputs[merged][worker_type] = list(pair)
+            else:
+                self._throughputs[merged][worker_type] = [pair[1], pair[0]]
+
+    def add_job(self, job: Job, timestamp=None) -> JobIdPair:
+        current_timestamp = self.get_current_timestamp()
+        job_id = JobIdPair(self._job_id_counter, None)
+        self._job_id_counter += 1
+        job.job_id = job_id
+        self._jobs[job_id] = job
+        self._steps_run_so_far[job_id] = {}
+        self._job_time_so_far[job_id] = {}
+        self._job_cost_so_far[job_id] = 0.0
+        self._job_timelines[job_id] = [[] for _ in range(job.scale_factor)]
+        self._throughputs[job_id] = {}
+        self._original_bs[job_id] = job.batch_size
+        self._original_num_steps[job_id] = job.total_steps
+        self._job_types[job_id] = job.job_type
+        self._num_jobs_in_trace += 1
+        self._num_failures_per_job[job_id] = 0
+        self._total_steps_run[job_id] = 0
+        self._cumulative_run_time[job_id] = {}
+        for worker_type in self._worker_types:
+            self._steps_run_so_far[job_id][worker_type] = 0
+            self._set_initial_throughput(job_id, worker_type)
+            if self._job_packing:
+                self._populate_job_combination_metadata(job_id, worker_type)
+            self._job_time_so_far[job_id][worker_type] = (
+                self._time_per_iteration / 2.0
+            )
+        self._per_job_latest_timestamps[job_id] = None
+        self._add_to_priorities(job_id)
+        self._need_to_update_allocation = True
+        self._bs_flags[job_id] = {"big_bs": False, "small_bs": False}
+        self._num_scheduled_rounds[job_id] = 0
+        self._num_queued_rounds[job_id] = 0
+        self._job_start_round[job_id[0]] = self._num_completed_rounds
+        self._steps_run_in_current_lease[job_id] = 0
+
+        if self.is_shockwave:
+            int_id = job_id[0]
+            metadata = JobMetadata(int_id, self._profiles[int_id], overclock=1.0)
+            metadata.register_submit(
+                self.get_current_timestamp()
+                if self._simulate
+                else self.get_current_timestamp() - self._start_timestamp
+            )
+            assert int_id not in self._throughput_timeline
+            self._throughput_timeline[int_id] = OrderedDict()
+            metadata.set_throughput_measurements(
+                self._throughput_timeline[int_id],
+                self._shockwave_planner.round_duration,
+            )
+            self._shockwave_planner.add_metadata(int_id, metadata)
+
+        if timestamp is None:
+            timestamp = current_timestamp
+        self._per_job_start_timestamps[job_id] = timestamp
+        logger.info("[Job dispatched] Job ID: %s duration: %s", job_id, job.duration)
+        return job_id
+
+    def _remove_job(self, job_id):
+        if isinstance(job_id, int):
+            job_id = JobIdPair(job_id, None)
+        self._completed_jobs.add(job_id)
+        duration = (
+            self._per_job_latest_timestamps[job_id]
+            - self._per_job_start_timestamps[job_id]
+        )
+        self._job_priority_weights[job_id] = self._jobs[job_id].priority_weight
+        del self._jobs[job_id]
+        self._job_completion_times[job_id] = duration
+        del self._steps_run_so_far[job_id]
+        del self._job_time_so_far[job_id]
+        del self._throughputs[job_id]
+        del self._num_failures_per_job[job_id]
+        self._job_end_round[job_id[0]] = self._num_completed_rounds
+        self._in_progress_updates.pop(job_id, None)
+        self._lease_update_requests.pop(job_id, None)
+        self._max_steps.pop(job_id, None)
+        self._jobs_with_extended_lease.discard(job_id)
+        self._steps_run_in_current_lease.pop(job_id, None)
+        if self.is_shockwave:
+            planner = self._shockwave_planner
+            if job_id[0] in planner.metadata:
+                planner.schedule_progress(
+                    job_id[0], planner.metadata[job_id[0]].epochs
+                )
+        if self._job_packing:
+            for other in [
+                o
+                for o in list(self._throughputs)
+                if isinstance(o, JobIdPair)
+                and o.is_pair()
+                and job_id.overlaps_with(o)
+            ]:
+                del self._throughputs[other]
+                self._job_time_so_far.pop(other, None)
+        self._remove_from_priorities(job_id)
+        self._need_to_update_allocation = True
+        logger.info("Remaining active jobs: %d", len(self._jobs))
+
+    def _get_remaining_steps(self, job_id):
+        return self._jobs[job_id].total_steps - self._total_steps_run[job_id]
+
+    # ------------------------------------------------------------------
+    # Priorities / allocation (reference :3498-3735)
+    # ------------------------------------------------------------------
+
+    def _add_to_priorities(self, job_id):
+        for worker_type in self._worker_types:
+            self._priorities[worker_type][job_id] = 0.0
+            self._deficits[worker_type][job_id] = 0.0
+            for other in self._throughputs:
+                if (
+                    isinstance(other, JobIdPair)
+                    and other.is_pair()
+                    and job_id.overlaps_with(other)
+                ):
+                    self._priorities[worker_type][other] = 0.0
+                    self._deficits[worker_type][other] = 0.0
+
+    def _remove_from_priorities(self, job_id):
+        for worker_type in self._worker_types:
+            for other in [
+                o
+                for o in list(self._priorities[worker_type])
+                if job_id.overlaps_with(o)
+            ]:
+                del self._priorities[worker_type][other]
+                del self._deficits[worker_type][other]
+
+    def _reset_time_run_so_far(self):
+        current_time = self.get_current_timestamp()
+        elapsed = current_time - self._last_reset_time
+        for worker_type in self._worker_types:
+            self._worker_time_so_far[worker_type] = 0.0
+            for job_id in self._job_time_so_far:
+                time_received = self._job_time_so_far[job_id].get(
+                    worker_type, self._time_per_iteration / 2.0
+                ) - (self._time_per_iteration / 2.0)
+                if job_id not in self._allocation:
+                    should_have = 0.0
+                else:
+                    should_have = self._allocation[job_id][worker_type] * elapsed
+                self._deficits[worker_type].setdefault(job_id, 0.0)
+                self._deficits[worker_type][job_id] += should_have - time_received
+                self._job_time_so_far[job_id][worker_type] = (
+                    self._time_per_iteration / 2.0
+                )
+                self._worker_time_so_far[worker_type] += (
+                    self._time_per_iteration / 2.0
+                )
+        self._last_reset_time = current_time
+        self._allocation_changed_since_last_time_reset = False
+
+    def _get_allocation_state(self):
+        state = {}
+        state["scale_factors"] = {
+            jid: self._jobs[jid].scale_factor for jid in self._jobs
+        }
+        state["priority_weights"] = {
+            jid: self._jobs[jid].priority_weight for jid in self._jobs
+        }
+        state["num_steps_remaining"] = {
+            jid: self._get_remaining_steps(jid)
+            - self._steps_run_in_current_lease.get(jid, 0)
+            for jid in self._jobs
+        }
+        state["times_since_start"] = {
+            jid: self.get_current_timestamp() - self._per_job_start_timestamps[jid]
+            for jid in self._jobs
+        }
+        state["throughputs"] = copy.deepcopy(self._throughputs)
+        state["per_round_schedule"] = self._per_round_schedule
+        state["cluster_spec"] = copy.deepcopy(self._cluster_spec)
+        return state
+
+    def _compute_allocation(self, state=None):
+        if state is None:
+            state = self._get_allocation_state()
+        name = getattr(self._policy, "name", "")
+        throughputs = state["throughputs"]
+        scale_factors = state["scale_factors"]
+        cluster_spec = state["cluster_spec"]
+        if name == "Shockwave" or name.lower() == "shockwave":
+            return {}
+        if name == "AlloX_Perf":
+            allocation = self._policy.get_allocation(
+                throughputs,
+                scale_factors,
+                state["times_since_start"],
+                state["num_steps_remaining"],
+                state["per_round_schedule"],
+                cluster_spec,
+            )
+        elif name.startswith("FinishTimeFairness"):
+            allocation = self._policy.get_allocation(
+                throughputs,
+                scale_factors,
+                state["priority_weights"],
+                state["times_since_start"],
+                state["num_steps_remaining"],
+                cluster_spec,
+            )
+        elif name.startswith("Isolated"):
+            allocation = self._policy.get_allocation(
+                throughputs, scale_factors, cluster_spec
+            )
+        elif name.startswith("MaxMinFairness"):
+            allocation = self._policy.get_allocation(
+                throughputs, scale_factors, state["priority_weights"], cluster_spec
+            )
+        elif name.startswith("MinTotalDuration"):
+            allocation = self._policy.get_allocation(
+                throughputs, scale_factors, state["num_steps_remaining"], cluster_spec
+            )
+        elif name.startswith("ThroughputNormalizedByCostSum") or name.startswith(
+            "ThroughputSum"
+        ):
+            allocation = self._policy.get_allocation(
+                throughputs, scale_factors, cluster_spec
+            )
+        elif name == "Proportional":
+            allocation = self._policy.get_allocation(throughputs, cluster_spec)
+        else:
+            allocation = self._policy.get_allocation(
+                throughputs, scale_factors, cluster_spec
+            )
+        return allocation or {}
+
+    def _update_priorities(self):
+        current_time = self.get_current_timestamp()
+        time_since_reset = current_time - self._last_reset_time
+        need_reset = (
+            time_since_reset >= self._minimum_time_between_allocation_resets
+            or self._last_reset_time == 0
+        )
+        if self._simulate:
+            need_reset = self._need_to_update_allocation and need_reset
+        else:
+            need_reset = (
+                self._allocation_changed_since_last_time_reset and need_reset
+            )
+        if need_reset:
+            self._reset_time_run_so_far()
+            if self._simulate:
+                self._allocation = self._compute_allocation()
+                self._need_to_update_allocation = False
+
+        fractions = {}
+        for worker_type in self._worker_types:
+            fractions[worker_type] = {}
+            worker_time = self._worker_time_so_far[worker_type]
+            for job_id in self._job_time_so_far:
+                if worker_time == 0.0 or worker_type not in self._job_time_so_far[job_id]:
+                    fractions[worker_type][job_id] = 0.0
+                else:
+                    fractions[worker_type][job_id] = (
+                        self._job_time_so_far[job_id][worker_type] / worker_time
+                    )
+            for job_id in self._priorities[worker_type]:
+                if job_id not in self._allocation:
+                    self._priorities[worker_type][job_id] = 0.0
+                    continue
+                alloc = self._allocation[job_id][worker_type]
+                new_priority = alloc * 1e9
+                tput = self._throughputs.get(job_id, {}).get(worker_type, 0)
+                zero_tput = (
+                    (job_id.is_pair() and (tput[0] == 0 or tput[1] == 0))
+                    if job_id.is_pair()
+                    else tput == 0
+                )
+                if alloc == 0.0:
+                    new_priority = 0.0
+                elif zero_tput:
+                    new_priority = 0.0
+                elif fractions[worker_type][job_id] > 0.0:
+                    new_priority = alloc / fractions[worker_type][job_id]
+                self._priorities[worker_type][job_id] = new_priority
+
+    # ------------------------------------------------------------------
+    # Round scheduling (reference :1049-1465)
+    # ------------------------------------------------------------------
+
+    def _schedule_jobs_on_workers_helper(self, worker_types):
+        if self.is_shockwave:
+            scheduled_jobs = {self._worker_type: []}
+            job_ids = self._shockwave_planner.round_schedule()
+            self._scheduled_jobs_in_prev_round = (
+                self._scheduled_jobs_in_current_round
+            )
+            self._scheduled_jobs_in_current_round = job_ids
+            for int_id in job_ids:
+                jid = JobIdPair(int_id, None)
+                if jid not in self._jobs:
+                    logger.warning(
+                        "job %s completed but still in round_schedule", int_id
+                    )
+                    continue
+                scheduled_jobs[self._worker_type].append(
+                    (jid, self._jobs[jid].scale_factor)
+                )
+            return scheduled_jobs
+
+        already_scheduled = set()
+        scheduled_jobs = {wt: [] for wt in worker_types}
+        num_workers_left = {wt: self._cluster_spec[wt] for wt in worker_types}
+        queue = []
+        for worker_type in worker_types:
+            entries = []
+            for job_id in self._priorities[worker_type]:
+                alloc = 0.0
+                if self._allocation and job_id in self._allocation:
+                    alloc = self._allocation[job_id][worker_type]
+                entries.append(
+                    (
+                        job_id,
+                        worker_type,
+                        self._priorities[worker_type][job_id],
+                        self._deficits[worker_type][job_id],
+                        alloc,
+                    )
+                )
+            queue += sorted(entries, key=lambda x: (x[2], x[3], x[4]), reverse=True)
+
+        name = getattr(self._policy, "name", "")
+        for job_id, worker_type, priority, _, _ in queue:
+            if num_workers_left[worker_type] == 0:
+                continue
+            singles = job_id.singletons()
+            if any(s in already_scheduled for s in singles):
+                continue
+            tput = self._throughputs[job_id][worker_type]
+            if job_id.is_pair():
+                if tput[0] <= 0 or tput[1] <= 0:
+                    continue
+                sf0 = self._jobs[singles[0]].scale_factor
+                sf1 = self._jobs[singles[1]].scale_factor
+                if sf0 != sf1:
+                    continue
+                scale_factor = sf0
+            else:
+                if tput <= 0:
+                    continue
+                scale_factor = self._jobs[job_id].scale_factor
+            if name.startswith("FIFO") and priority <= 0.0:
+                continue
+            if scale_factor > num_workers_left[worker_type]:
+                continue
+            num_workers_left[worker_type] -= scale_factor
+            for s in singles:
+                already_scheduled.add(s)
+            scheduled_jobs[worker_type].append((job_id, scale_factor))
+        return scheduled_jobs
+
+    def _assign_workers_to_job(
+        self, job_id, scale_factor, worker_type, worker_state, worker_assignments
+    ):
+        worker_ids = worker_state["worker_ids"]
+        assigned = worker_state["assigned_worker_ids"]
+        ptr = worker_state["server_id_ptr"]
+        ids_for_job = list(worker_assignments.get(job_id, ()))
+        while len(ids_for_job) < scale_factor and ptr < len(worker_ids):
+            if not worker_ids[ptr]:
+                ptr += 1
+                continue
+            wid = worker_ids[ptr][0]
+            if wid not in assigned:
+                ids_for_job.append(wid)
+                assigned.add(wid)
+            worker_ids[ptr].pop(0)
+        if len(ids_for_job) != scale_factor:
+            raise RuntimeError(f"could not assign workers to job {job_id}")
+        worker_assignments[job_id] = tuple(ids_for_job)
+        worker_state["server_id_ptr"] = ptr
+        for single in job_id.singletons():
+            if self._simulate:
+                self._per_job_latest_timestamps[single] = (
+                    self.get_current_timestamp()
+                )
+                self._running_jobs.add(single)
+
+    def _schedule_jobs_on_workers(self):
+        if not self.is_shockwave:
+            self._update_priorities()
+
+        worker_types = [
+            wt
+            for wt in [self._worker_type, "v100", "p100", "k80"]
+            if wt in self._worker_type_to_worker_id_mapping
+        ]
+        worker_types = list(dict.fromkeys(worker_types))
+        name = getattr(self._policy, "name", "")
+        if "Perf" not in name and "Packing" not in name:
+            self._worker_type_shuffler.shuffle(worker_types)
+
+        new_assignments = OrderedDict()
+        scheduled_jobs = self._schedule_jobs_on_workers_helper(worker_types)
+
+        worker_state = {}
+        for worker_type in worker_types:
+            scheduled_jobs.setdefault(worker_type, [])
+            scheduled_jobs[worker_type].sort(key=lambda x: x[1], reverse=True)
+            worker_state[worker_type] = {
+                "worker_ids": copy.deepcopy(
+                    self._worker_type_to_worker_id_mapping[worker_type]
+                ),
+                "assigned_worker_ids": set(),
+                "server_id_ptr": 0,
+            }
+
+        prev_worker_types = {
+            job_id: self._worker_id_to_worker_type_mapping[wids[0]]
+            for job_id, wids in self._current_worker_assignments.items()
+        }
+
+        for worker_type in worker_types:
+            state = worker_state[worker_type]
+            assigned = state["assigned_worker_ids"]
+            scale_factors = sorted(
+                {sf for _, sf in scheduled_jobs[worker_type]}, reverse=True
+            )
+            for current_sf in scale_factors:
+                # keep jobs on their current workers when possible
+                for job_id, sf in scheduled_jobs[worker_type]:
+                    if sf != current_sf:
+                        continue
+                    if prev_worker_types.get(job_id) == worker_type:
+                        prev_ids = self._current_worker_assignments[job_id]
+                        if all(w not in assigned for w in prev_ids):
+                            new_assignments[job_id] = prev_ids
+                            assigned.update(prev_ids)
+                for job_id, sf in scheduled_jobs[worker_type]:
+                    if sf != current_sf:
+                        continue
+                    if not self.is_shockwave and job_id not in self._allocation:
+                        continue
+                    if job_id in new_assignments:
+                        continue
+                    self._assign_workers_to_job(
+                        job_id, sf, worker_type, state, new_assignments
+                    )
+                    if self.is_shockwave:
+                        self._allocation[job_id] = {self._worker_type: -1.0}
+
+        counts = collections.Counter(
+            w for wids in new_assignments.values() for w in wids
+        )
+        for wid, cnt in counts.items():
+            if cnt != 1:
+                raise RuntimeError(f"worker {wid} assigned {cnt} times")
+
+        assignments = {
+            job_id[0]: wids for job_id, wids in new_assignments.items()
+        }
+        self._per_round_schedule.append(assignments)
+        self._num_jobs_in_curr_round.append(len(self._jobs))
+        for job_id in self._jobs:
+            if job_id[0] in assignments:
+                self._num_scheduled_rounds[job_id] += 1
+            else:
+                self._num_queued_rounds[job_id] += 1
+        return new_assignments
+
+    # ------------------------------------------------------------------
+    # Step accounting (reference :1425-1516)
+    # ------------------------------------------------------------------
+
+    def _get_num_steps(self, job_id, worker_type, single_job_id=None):
+        if self._simulate and job_id.is_pair():
+            assert single_job_id is not None
+            index = 0 if job_id.singletons()[0] == single_job_id else 1
+            num_steps = int(
+                self._throughputs[job_id][worker_type][index]
+                * self._time_per_iteration
+            )
+        else:
+            tput = self._throughputs[job_id][worker_type]
+            if job_id.is_pair():
+                index = 0 if job_id.singletons()[0] == single_job_id else 1
+                tput = tput[index]
+            num_steps = int(tput * self._time_per_iteration)
+        target = single_job_id if single_job_id is not None else job_id
+        return min(num_steps, self._get_remaining_steps(target))
+
+    def _get_job_steps_and_finish_times(self, job_id, worker_type):
+        max_finish_time = self.get_current_timestamp()
+        all_num_steps = []
+        for single in job_id.singletons():
+            num_steps = self._get_num_steps(job_id, worker_type, single)
+            all_num_steps.append(num_steps)
+            tput = self._throughputs[job_id][worker_type]
+            if job_id.is_pair():
+                index = 0 if job_id.singletons()[0] == single else 1
+                tput = tput[index]
+            if tput <= 0:
+                raise RuntimeError(
+                    f"throughput for job {single} on {worker_type} <= 0"
+                )
+            finish_time = self.get_current_timestamp() + num_steps / tput
+            max_finish_time = max(max_finish_time, finish_time)
+            self._running_jobs.add(single)
+        return all_num_steps, max_finish_time
+
+    def _update_throughput(self, job_id, worker_type, all_num_steps, all_execution_times):
+        if job_id not in self._throughputs:
+            return
+        for i, single in enumerate(job_id.singletons()):
+            int_id = single[0]
+            self._throughput_timeline.setdefault(int_id, OrderedDict())
+            tput = (
+                all_num_steps[i] / all_execution_times[i]
+                if all_execution_times[i] > 0
+                else 0.0
+            )
+            bs = self._jobs[single].batch_size if single in self._jobs else 0
+            self._throughput_timeline[int_id][self._num_completed_rounds] = (
+                tput,
+                bs,
+            )
+        if not self._simulate:
+            # EMA between old value and new measurement (reference :596-601)
+            for i, single in enumerate(job_id.singletons()):
+                if all_execution_times[i] <= 0:
+                    continue
+                new_tput = all_num_steps[i] / all_execution_times[i]
+                if job_id.is_pair():
+                    old = self._throughputs[job_id][worker_type][i]
+                    self._throughputs[job_id][worker_type][i] = (
+                        EMA_ALPHA * new_tput + (1 - EMA_ALPHA) * old
+                    )
+                else:
+                    old = self._throughputs[job_id][worker_type]
+                    self._throughputs[job_id][worker_type] = (
+                        EMA_ALPHA * new_tput + (1 - EMA_ALPHA) * old
+                    )
+
+    # ------------------------------------------------------------------
+    # Batch-size adaptation (reference :1604-1726, :4731-4931)
+    # ------------------------------------------------------------------
+
+    def _get_num_epochs(self, job_type, batch_size, num_steps):
+        model = job_type[: job_type.find(" ")]
+        spe = datasets.steps_per_epoch(model, batch_size)
+        return math.ceil(num_steps / spe)
+
+    _MAX_BS = {"LM": 80, "ResNet-18": 256, "ResNet-50": 128, "Recommendation": 8192}
+    _MIN_BS = {"LM": 5, "ResNet-18": 16, "ResNet-50": 16, "Transformer": 16,
+               "Recommendation": 512}
+
+    def _simulate_gns(self, job_id):
+        from ..core import bs_patterns
+
+        job = self._jobs[job_id]
+        model = job.model
+        batch_size = job.batch_size
+        original_bs = self._original_bs[job_id]
+        total_steps_run = self._total_steps_run[job_id]
+        current_epoch = self._get_num_epochs(job.job_type, batch_size, total_steps_run)
+        pattern = bs_patterns.gns_bs_pattern(
+            job.job_type,
+            original_bs,
+            max(760, current_epoch + 2),
+            job.scale_factor,
+        )
+        if pattern[current_epoch + 1] > batch_size or pattern[current_epoch] > batch_size:
+            if self._MAX_BS.get(model) != batch_size:
+                self._bs_flags[job_id]["big_bs"] = True
+
+    def _simulate_accordion(self, job_id):
+        job = self._jobs[job_id]
+        model = job.model
+        batch_size = job.batch_size
+        original_bs = self._original_bs[job_id]
+        total_steps_run = self._total_steps_run[job_id]
+        current_epoch = self._get_num_epochs(job.job_type, batch_size, total_steps_run)
+
+        if model == "Transformer":
+            return
+        if model == "LM":
+            in_cr = current_epoch < 10
+        elif model == "Recommendation":
+            if original_bs in (512, 1024):
+                in_cr = current_epoch < 30
+            elif original_bs == 2048:
+                in_cr = current_epoch < 40
+            else:
+                in_cr = current_epoch < 10
+        elif model == "ResNet-50":
+            in_cr = (current_epoch % 30) < 10
+        elif model == "ResNet-18":
+            head = 20 if original_bs == 256 else 10
+            in_cr = (
+                current_epoch < head
+                or 150 <= current_epoch < 160
+                or 250 <= current_epoch < 260
+            )
+        else:
+            return
+        if batch_size == original_bs and not in_cr:
+            if self._MAX_BS.get(model) != batch_size:
+                self._bs_flags[job_id]["big_bs"] = True
+        elif batch_size != original_bs and in_cr:
+            if self._MIN_BS.get(model) != batch_size:
+                self._bs_flags[job_id]["small_bs"] = True
+
+    def _scale_bs_and_iters(self, job_id):
+        """Apply a pending batch-size change, preserving epoch counts
+        (reference :4731-4931)."""
+        if job_id is None:
+            return
+        if isinstance(job_id, int):
+            job_id = JobIdPair(job_id, None)
+        flags = self._bs_flags.get(job_id)
+        if not flags or not (flags["big_bs"] or flags["small_bs"]):
+            return
+        job = self._jobs[job_id]
+        old_bs = job.batch_size
+        model = job.model
+        mode = job.mode
+        original_bs = self._original_bs[job_id]
+
+        if model in self._MAX_BS and original_bs == self._MAX_BS[model]:
+            flags["big_bs"] = flags["small_bs"] = False
+            return
+        if mode == "gns":
+            assert flags["big_bs"]
+            new_bs = 2 * old_bs
+        elif mode == "accordion":
+            new_bs = self._MAX_BS[model] if flags["big_bs"] else original_bs
+        else:
+            new_bs = old_bs
+
+        job.update_bs(new_bs)
+        for worker_type in self._worker_types:
+            key = (job.job_type, job.scale_factor)
+            if key not in self._oracle_throughputs[worker_type]:
+                logger.error(
+                    "job %s requested unprofiled bs %s; reverting", job_id, key
+                )
+                flags["big_bs"] = flags["small_bs"] = False
+                job.update_bs(old_bs)
+                return
+            self._throughputs[job_id][worker_type] = self._oracle_throughputs[
+                worker_type
+            ][key]["null"]
+
+        # preserve epoch counts across the rescale
+        spe_old = datasets.steps_per_epoch(model, old_bs)
+        spe_new = datasets.steps_per_epoch(model, new_bs)
+        total_steps = job.total_steps
+        old_total_epochs = math.ceil(total_steps / spe_old)
+        new_total_steps = math.ceil(total_steps * old_bs / new_bs)
+        if math.ceil(new_total_steps / spe_new) != old_total_epochs:
+            new_total_steps = spe_new * old_total_epochs
+        job.total_steps = new_total_steps
+
+        total_steps_run = self._total_steps_run[job_id]
+        completed_epochs = math.ceil(total_steps_run / spe_old)
+        new_steps_run = completed_epochs * spe_new
+        self._total_steps_run[job_id] = new_steps_run
+        for wt in self._steps_run_so_far[job_id]:
+            self._steps_run_so_far[job_id][wt] = new_steps_run
+
+        flags["big_bs"] = flags["small_bs"] = False
+
+    # ------------------------------------------------------------------
+    # Done callback (simulation path of reference :4341-4729)
+    # ------------------------------------------------------------------
+
+    def _done_callback(
+        self, job_id, worker_id, all_num_steps, all_execution_times,
+        all_iterator_logs=None,
+    ):
+        to_remove = []
+        self._cumulative_run_time[job_id].setdefault(worker_id, 0.0)
+        self._cumulative_run_time[job_id][worker_id] += float(
+            np.max(all_execution_times)
+        )
+
+        if job_id in self._jobs:
+            run_time_so_far = (
+                sum(self._cumulative_run_time[job_id].values())
+                / self._jobs[job_id].scale_factor
+            )
+            is_over_deadline = run_time_so_far > int(
+                self._jobs[job_id].duration * 1.5
+            )
+        else:
+            is_over_deadline = True
+
+        is_active = {s: s in self._jobs for s in job_id.singletons()}
+        if not any(is_active.values()):
+            return
+
+        worker_type = self._worker_id_to_worker_type_mapping[worker_id]
+        scale_factor = len(self._current_worker_assignments.get(job_id, (0,)))
+        self._in_progress_updates[job_id].append(
+            (worker_id, all_num_steps, all_execution_times, all_iterator_logs)
+        )
+        if len(self._in_progress_updates[job_id]) < scale_factor:
+            return
+        self._in_progress_updates[job_id].sort(key=lambda x: x[0])
+
+        micro_task_succeeded = True
+        agg_num_steps = [0] * len(job_id.singletons())
+        agg_execution_times = [0.0] * len(job_id.singletons())
+        for i, update in enumerate(self._in_progress_updates[job_id]):
+            _, steps_, times_, logs_ = update
+            for j, single in enumerate(job_id.singletons()):
+                if not is_active[single]:
+                    continue
+                if steps_[j] <= 0 and times_[j] <= 0:
+                    micro_task_succeeded = False
+                    break
+            for j, single in enumerate(job_id.singletons()):
+                agg_num_steps[j] += steps_[j]
+                agg_execution_times[j] = max(agg_execution_times[j], times_[j])
+                if logs_ is not None:
+                    self._job_timelines[single][i].extend(logs_[j].split("\n"))
+        all_worker_ids = sorted(
+            u[0] for u in self._in_progress_updates[job_id]
+        )
+        self._in_progress_updates[job_id] = []
+        for single in job_id.singletons():
+            self._lease_update_requests[single] = []
+            self._max_steps[single] = None
+
+        if not micro_task_succeeded:
+            logger.info("[Micro-task failed] Job ID: %s", job_id)
+            if not job_id.is_pair() and is_active[job_id]:
+                self._num_failures_per_job[job_id] += 1
+                if self._num_failures_per_job[job_id] >= MAX_FAILED_ATTEMPTS:
+                    logger.info("[Job failed] Job ID: %s", job_id)
+                    to_remove.append(job_id)
+            self._need_to_update_allocation = True
+        else:
+            self._num_failures_per_job[job_id] = 0
+            for single, num_steps, execution_time in zip(
+                job_id.singletons(), agg_num_steps, agg_execution_times
+            ):
+                if not is_active[single]:
+                    continue
+                if single in self._running_jobs:
+                    self._running_jobs.remove(single)
+                    self._steps_run_so_far[single][worker_type] += num_steps
+                    self._total_steps_run[single] += num_steps
+                    self._steps_run_in_current_lease[single] = 0
+                    remaining = self._get_remaining_steps(single)
+                    if remaining <= 0 or is_over_deadline:
+                        start = self._per_job_start_timestamps[single]
+                        finish = self._per_job_latest_timestamps[single]
+                        logger.info(
+                            "[Job succeeded] Job ID: %s duration %.1f",
+                            single,
+                            finish - start,
+                        )
+                        to_remove.append(single)
+            max_exec = float(np.max(agg_execution_times))
+            if job_id in self._job_time_so_far:
+                self._job_time_so_far[job_id][worker_type] += max_exec
+                self._worker_time_so_far[worker_type] += max_exec
+            for wid in all_worker_ids:
+                self._cumulative_worker_time_so_far[wid] += max_exec
+
+        self._update_throughput(
+            job_id, worker_type, agg_num_steps, agg_execution_times
+        )
+
+        for single in job_id.singletons():
+            self._scale_bs_and_iters(single)
+
+        for single in to_remove:
+            self._remove_job(single)
+            if self.is_shockwave and single[0] in self._shockwave_planner.metadata:
+                self._shockwave_planner.remove_metadata(single[0])
+                self._shockwave_job_completed_flag = True
+
+        for single in job_id.singletons():
+            flags = self._bs_flags.get(single)
+            if flags and (flags["big_bs"] or flags["small_bs"]):
+                self._need_to_update_allocation = True
+                flags["big_bs"] = False
+                flags["small_bs"] = False
+
+    # ------------------------------------------------------------------
+    # Shockwave per-round update (reference :2270-2380)
+    # ------------------------------------------------------------------
+
+    def _update_shockwave_planner(self, jobs_with_extended_lease=None):
+        planner = self._shockwave_planner
+        scheduled = (
+            self._scheduled_jobs_in_current_round
+            if self._simulate
+            else self._scheduled_jobs_in_prev_round
+        )
+        for int_id in scheduled:
+            jid = JobIdPair(int_id, None)
+            if jid in self._completed_jobs:
+                if int_id in planner.metadata:
+                    planner.schedule_progress(
+                        int_id, planner.metadata[int_id].epochs
+                    )
+                continue
+            if jid not in self._jobs:
+                continue
+            steps_run = self._steps_run_so_far.get(jid, {}).get(
+                self._worker_type, 0
+            )
+            if not self._simulate and jobs_with_extended_lease:
+                if jid in jobs_with_extended_lease:
+                    steps_run += self._steps_run_in_current_lease.get(jid, 0)
+            bs = self._jobs[jid].batch_size
+            spe = datasets.steps_per_epoch(self._jobs[jid].model, bs)
+            current_epoch = math.floor(steps_run / spe)
+            if int_id in planner.metadata:
+                planner.schedule_progress(int_id, min(
+                    current_epoch, planner.metadata[int_id].epochs))
+
+        all_ids = {j[0] for j in self._jobs}
+        for int_id in all_ids - set(scheduled):
+            planner.deschedule_waiting_delay(int_id, self._time_per_iteration)
+
+        planner.increment_round_ptr()
+        self._iround_reopt += 1
+        if self._shockwave_job_completed_flag or self._iround_reopt >= REOPT_ROUNDS:
+            self._shockwave_job_completed_flag = False
+            self._iround_reopt = 0
+            planner.set_resolve()
+
+    # ------------------------------------------------------------------
+    # Simulation loop (reference simulate() :1728-2268)
+    # ------------------------------------------------------------------
+
+    def simulate(self, cluster_spec, arrival_times, jobs,
+                 num_gpus_per_server=None, debug=False):
+        queued_jobs = list(zip(arrival_times, jobs))
+        remaining_jobs = len(jobs)
+        running_jobs = []  # heap of (-finish_time, job_id, worker_ids, steps)
+        current_round_start_time = 0
+        current_round_end_time = None
+
+        for worker_type in sorted(cluster_spec):
+            num_gpus = 1
+            if num_gpus_per_server:
+                num_gpus = num_gpus_per_server[worker_type]
+            for _ in range(cluster_spec[worker_type] // num_gpus):
+                self.register_worker(worker_type, num_gpus=num_gpus)
+
+        if queued_jobs:
+            self._current_timestamp = queued_jobs[0][0]
+        current_round = 0
+
+        while True:
+            if remaining_jobs == 0:
+                break
+            next_job_arrival_time = queued_jobs[0][0] if queued_jobs else None
+
+            # advance the clock
+            max_timestamp = 0
+            if running_jobs and -running_jobs[0][0] > max_timestamp:
+                max_timestamp = -running_jobs[0][0]
+                if current_round_end_time is not None:
+                    current_round_start_time = current_round_end_time
+                current_round_end_time = max_timestamp
+            if max_timestamp > 0:
+                self._current_timestamp = max_timestamp
+            else:
+                self._current_timestamp = next_job_arrival_time
+
+            # drain completed micro-tasks
+            while running_jobs:
+                finish_time, job_id, worker_ids, all_num_steps = running_jobs[0]
+                finish_time = -finish_time
+                if finish_time > self._current_timestamp:
+                    break
+                all_execution_times = []
+                nfs_slowdown_factor = 1
+                for single in job_id.singletons():
+                    execution_time = finish_time - current_round_start_time
+                    if current_round != 1 and current_round >= 2:
+                        prev_sched = self._per_round_schedule[current_round - 2]
+                        if single[0] not in prev_sched:
+                            # migration: inject checkpoint/restore overhead
+                            if (
+                                execution_time != 0
+                                and self._time_per_iteration - 5 < execution_time
+                            ):
+                                nfs_slowdown_factor = (
+                                    execution_time - PREEMPTION_OVERHEAD_S
+                                ) / execution_time
+                                execution_time -= PREEMPTION_OVERHEAD_S
+                    all_execution_times.append(execution_time)
+                    self._per_job_latest_timestamps[single] = finish_time
+                self._in_progress_updates[job_id] = []
+                scale_factor = self._jobs[job_id.singletons()[0]].scale_factor
+                all_num_steps = [
+                    int(x * nfs_slowdown_factor) for x in all_num_steps
+                ]
+                total_steps = [0] * len(job_id.singletons())
+                for i, worker_id in enumerate(worker_ids):
+                    if i == len(worker_ids) - 1:
+                        steps_i = [
+                            all_num_steps[j] - total_steps[j]
+                            for j in range(len(all_num_steps))
+                        ]
+                    else:
+                        steps_i = [x // scale_factor for x in all_num_steps]
+                    for j in range(len(steps_i)):
+                        total_steps[j] += steps_i[j]
+                    self._done_callback(
+                        job_id, worker_id, steps_i, all_execution_times
+                    )
+                for single in job_id.singletons():
+                    if single not in self._jobs:
+                        remaining_jobs -= 1
+                heapq.heappop(running_jobs)
+
+            # dynamic adaptation twins
+            for jid in list(self._jobs.keys()):
+                if self._jobs[jid].mode == "accordion":
+                    self._simulate_accordion(jid)
+                elif self._jobs[jid].mode == "gns":
+                    self._simulate_gns(jid)
+
+            if self.is_shockwave and self._current_timestamp != 0.0:
+                self._update_shockwave_planner()
+
+            assert len(running_jobs) == 0
+
+            # admit newly arrived jobs
+            while queued_jobs:
+                arrival_time, job = queued_jobs[0]
+                if arrival_time > self._current_timestamp:
+                    break
+                self.add_job(job, timestamp=arrival_time)
+                queued_jobs.pop(0)
+
+            if len(self._jobs) == 0:
+                logger.warning("simulation complete: no jobs left")
+                break
+
+            # schedule the round
+            scheduled_jobs = self._schedule_jobs_on_workers()
+            if not scheduled_jobs and self._jobs and not queued_jobs:
+                # stale allocation can schedule nothing while jobs remain
+                # (e.g. sticky FIFO between allocation resets): force a
+                # recompute so the clock can advance
+                self._need_to_update_allocation = True
+                self._last_reset_time = -self._minimum_time_between_allocation_resets
+                scheduled_jobs = self._schedule_jobs_on_workers()
+                if not scheduled_jobs:
+                    raise RuntimeError(
+                        "no jobs schedulable while %d jobs active"
+                        % len(self._jobs)
+                    )
+            for job_id in self._current_worker_assignments:
+                if any(x in self._jobs for x in job_id.singletons()):
+                    self._num_lease_extension_opportunities += 1
+            for job_id in scheduled_jobs:
+                if job_id in self._current_worker_assignments:
+                    if set(self._current_worker_assignments[job_id]) == set(
+                        scheduled_jobs[job_id]
+                    ):
+                        self._num_lease_extensions += 1
+            self._current_worker_assignments = scheduled_jobs
+
+            for job_id, worker_ids in scheduled_jobs.items():
+                worker_type = self._worker_id_to_worker_type_mapping[
+                    worker_ids[0]
+                ]
+                all_num_steps, max_finish_time = (
+                    self._get_job_steps_and_finish_times(job_id, worker_type)
+                )
+                heapq.heappush(
+                    running_jobs,
+                    (-max_finish_time, job_id, worker_ids, all_num_steps),
+                )
+
+            current_round += 1
+            self._num_completed_rounds += 1
+            if self._max_rounds is not None and current_round >= self._max_rounds:
+                break
+
+        logger.info(
+            "Total duration/makespan: %.3f s (%.2f h)",
+            self._current_timestamp,
+            self._current_timestamp / 3600.0,
+        )
+        return self._current_timestamp
+
+    # ------------------------------------------------------------------
+    # Metrics (reference :2779-3107)
+    # ------------------------------------------------------------------
+
+    def get_average_jct(self, job_ids=None):
+        if not self._job_completion_times:
+            return None
+        if job_ids is None:
+            job_ids = sorted(self._job_completion_times.keys())
+        cts = [
+            self._job_completion_times[j]
+            for j in job_ids
+            if self._job_completion_times.get(j) is not None
+        ]
+        return (
+            float(np.mean(cts)),
+            float(scipy.stats.mstats.gmean(cts)),
+            float(scipy.stats.hmean(cts)),
+            cts,
+        )
+
+    def get_finish_time_fairness(self, job_ids=None):
+        """rho per job, two contention models (reference :2865-2964)."""
+        num_gpus = len(self._worker_ids)
+        if not self._job_completion_times:
+            return None
+        if job_ids is None:
+            job_ids = sorted(self._job_completion_times.keys())
+        static_list, themis_list = [], []
+        for job_id in job_ids:
+            ct = self._job_completion_times.get(job_id)
+            if ct is None:
+                continue
+            exclusive = sum(
+                self._profiles[job_id[0]]["duration_every_epoch"]
+            )
+            contention = max(1.0, self._num_jobs_in_trace / num_gpus)
+            static_list.append(round(ct / (exclusive * contention), 5))
+            start_round = self._job_start_round.get(job_id[0], 0)
+            end_round = self._job_end_round.get(
+                job_id[0], self._num_completed_rounds
+            )
+            if end_round > start_round:
+                mean_jobs = float(
+                    np.mean(self._num_jobs_in_curr_round[start_round:end_round])
+                )
+            else:
+                mean_jobs = float(self._num_jobs_in_trace)
+            contention = max(1.0, mean_jobs / num_gpus)
+            themis_list.append(round(ct / (exclusive * contention), 5))
+        return static_list, themis_list
+
+    def get_cluster_utilization(self):
+        utilizations = []
+        now = self.get_current_timestamp()
+        for wid, wtime in self._cumulative_worker_time_so_far.items():
+            total = now - self._worker_start_times[wid]
+            if total <= 0:
+                continue
+            utilizations.append(round(wtime / total, 5))
+        return (float(np.mean(utilizations)) if utilizations else 0.0, utilizations)
+
+    def get_envy_list(self):
+        """Pairwise envy: how much more time another job received relative
+        to own allocation-normalized share (reference :2966-3014 in spirit:
+        completed-job envy ratios over per-round scheduled counts)."""
+        envy_ratios = []
+        sched = self._num_scheduled_rounds
+        for a in self._job_completion_times:
+            ra = sched.get(a, 0)
+            for b in self._job_completion_times:
+                if a == b:
+                    continue
+                rb = sched.get(b, 0)
+                if ra > 0:
+                    envy_ratios.append(rb / ra)
+        return envy_ratios
+
+    def get_num_lease_extensions(self):
+        if self._num_lease_extension_opportunities > 0:
+            pct = (
+                100.0 * self._num_lease_extensions
+            ) / self._num_lease_extension_opportunities
+        else:
+            pct = 0.0
+        return (
+            pct,
+            self._num_lease_extensions,
+            self._num_lease_extension_opportunities,
+        )
+
+    def get_per_round_schedule(self):
+        return self._per_round_schedule
+
+    def get_job_completion_times(self):
+        return dict(self._job_completion_times)
+
+    def is_done(self, jobs_to_complete=None):
+        if jobs_to_complete is not None:
+            return jobs_to_complete.issubset(self._completed_jobs)
+        return len(self._jobs) == 0
