@@ -1,0 +1,240 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — driver contract entry point.
+
+Measures the framework's flagship training step: ResNet-18 on synthetic
+CIFAR-10 under the lease-preemptible iterator, with the fused CDNA4
+optimizer kernel and (for --gpus > 1) bucketed data parallelism over
+RCCL/xGMI — the per-GPU training path every scheduled job runs on this
+framework (BASELINE configs 2-5).
+
+Metric: aggregate samples/s across all ranks at per-GPU batch 16 — the
+one per-job throughput number BASELINE.md quotes for the reference
+(ResNet-18 bs16 = 57.68 steps/s = 922.9 samples/s on one V100).
+``vs_baseline`` is per-GPU-normalized: (value / n_gpus) / 922.9.
+
+Scheduler-level metrics (makespan + worst FTF rho on the 120-job trace,
+the reference's headline table) are produced by ``--trace-sim`` /
+scripts/simulate.py and recorded in profiles/.
+
+Launch (multi-GPU, by the driver):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+V100_RESNET18_BS16_STEPS_PER_S = 57.68  # BASELINE.md "Other reference numbers"
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=50)
+    p.add_argument("--batch-size", type=int, default=16, help="per-GPU batch")
+    p.add_argument("--model", default="resnet18")
+    p.add_argument("--no-graphs", action="store_true",
+                   help="disable hipGraph capture of the train step")
+    p.add_argument("--trace-sim", action="store_true",
+                   help="also run the 120-job scheduler simulation (rank 0)")
+    return p.parse_args()
+
+
+def build_flagship(args, device):
+    from shockwave_amd.models import resnet18_cifar
+    from shockwave_amd.ops.optim import FusedSGD
+    from shockwave_amd.parallel import BucketedDataParallel
+
+    model = resnet18_cifar().to(device)
+    if device.type == "cuda":
+        # NHWC keeps MIOpen on its native igemm path (no per-conv
+        # batched_transpose kernels — ~90 launches/step on this model)
+        model = model.to(memory_format=torch.channels_last)
+        torch.backends.cudnn.benchmark = True
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        model = BucketedDataParallel(model)
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=5e-4)
+    criterion = torch.nn.CrossEntropyLoss().to(device)
+    return model, opt, criterion
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world_size > 1
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if distributed:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+
+    if use_cuda:
+        import shockwave_amd.ops as ops
+
+        if not ops.HAVE_EXT:
+            raise RuntimeError(
+                "HIP extension not built — bench must run the native path"
+            )
+
+    model, opt, criterion = build_flagship(args, device)
+    from shockwave_amd.runtime.lease_iterator import LeaseIterator, NullLeaseClient
+    from shockwave_amd.workloads import common
+    from shockwave_amd.data.synthetic import SyntheticImages
+    from torch.utils.data import DataLoader
+
+    bs = args.batch_size
+    dataset = SyntheticImages(50000, image_size=32, num_classes=10, seed=rank)
+    loader = DataLoader(dataset, batch_size=bs, shuffle=False, num_workers=0,
+                        drop_last=True, pin_memory=use_cuda)
+    ckpt_dir = os.path.join("/tmp", f"swq_bench_{rank}")
+    lease_it = LeaseIterator(
+        loader, ckpt_dir, lambda: None, lambda s: None,
+        synthetic_data=False, write_on_close=False, client=NullLeaseClient(),
+    )
+    data_iter = iter(lease_it)
+
+    def next_batch():
+        nonlocal data_iter
+        try:
+            return next(data_iter)
+        except StopIteration:
+            data_iter = iter(lease_it)
+            return next(data_iter)
+
+    channels_last = use_cuda
+
+    def compute_step(x, y):
+        common.zero_grads(model)
+        loss = criterion(model(x), y)
+        loss.backward()
+        common.finish_sync(model)
+        opt.step()
+        return loss
+
+    graphed = None
+    if use_cuda and not args.no_graphs:
+        from shockwave_amd.parallel.graphs import try_graph_step
+
+        static_x = torch.zeros(bs, 3, 32, 32, device=device)
+        if channels_last:
+            static_x = static_x.to(memory_format=torch.channels_last)
+        static_y = torch.zeros(bs, dtype=torch.long, device=device)
+        graphed = try_graph_step(compute_step, [static_x, static_y])
+
+    def train_step():
+        x, y = next_batch()
+        x = x.to(device, non_blocking=True)
+        if channels_last:
+            x = x.to(memory_format=torch.channels_last)
+        y = y.to(device, non_blocking=True)
+        if graphed is not None:
+            graphed(x, y)
+        else:
+            compute_step(x, y)
+
+    def barrier_sync():
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    model.train()
+    for _ in range(args.warmup):
+        train_step()
+
+    barrier_sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        train_step()
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device=device if use_cuda else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world_size if use_cuda else args.gpus
+    steps_per_s = args.steps / elapsed
+    samples_per_s = steps_per_s * bs * world_size
+    ms_per_step = 1000.0 * elapsed / args.steps
+    baseline_samples = V100_RESNET18_BS16_STEPS_PER_S * 16
+    vs_baseline = (
+        (samples_per_s / world_size) / baseline_samples if bs == 16 else None
+    )
+
+    result = {
+        "metric": "samples_per_s",
+        "value": samples_per_s,
+        "unit": "samples/s",
+        "n_gpus": world_size,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": vs_baseline,
+        "dtype": "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "ResNet-18",
+            "global_batch": bs * world_size,
+            "seq_len": None,
+            "parallelism": f"dp{world_size}",
+            "per_gpu_batch": bs,
+            "baseline_ref": "V100 ResNet-18 bs16 57.68 steps/s (BASELINE.md)",
+            "framework_path": "LeaseIterator + FusedSGD(HIP) + "
+                              "BucketedDataParallel(RCCL)",
+            "hipgraph": graphed is not None,
+            "memory_format": "channels_last" if channels_last else "nchw",
+        },
+    }
+
+    if args.trace_sim and rank == 0:
+        sys.path.insert(0, os.path.join(os.path.dirname(__file__), "scripts"))
+        from simulate import run_simulation
+
+        sim = run_simulation(
+            "traces/tacc_like_120.trace",
+            "traces/mi355x_throughputs.json",
+            "shockwave",
+            num_gpus=32,
+        )
+        result["config"]["sim_makespan_h"] = sim["makespan_h"]
+        result["config"]["sim_worst_ftf_rho"] = sim["worst_ftf_rho"]
+
+    if rank == 0:
+        print(json.dumps(result))
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
