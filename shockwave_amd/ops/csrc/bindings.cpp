@@ -49,9 +49,15 @@ void check_lists(const std::vector<std::vector<torch::Tensor>>& lists) {
             TORCH_CHECK(ten.is_cuda(), "swq ops require device tensors");
             TORCH_CHECK(ten.scalar_type() == torch::kFloat32,
                         "swq ops are fp32-only (got ", ten.scalar_type(), ")");
-            TORCH_CHECK(ten.is_contiguous(), "tensor ", t, " not contiguous");
+            // flat elementwise kernels: any dense non-overlapping layout is
+            // fine (NCHW-contiguous or channels_last) as long as every list
+            // shares the layout per tensor index
+            TORCH_CHECK(ten.is_non_overlapping_and_dense(),
+                        "tensor ", t, " not dense");
             TORCH_CHECK(ten.numel() == lists[0][t].numel(),
                         "numel mismatch across lists at tensor ", t);
+            TORCH_CHECK(ten.strides() == lists[0][t].strides(),
+                        "stride/layout mismatch across lists at tensor ", t);
         }
     }
 }

@@ -85,7 +85,10 @@ class BucketedDataParallel(torch.nn.Module):
         offset = 0
         for p in params:
             n = p.numel()
-            p.grad = buffer[offset : offset + n].view_as(p)
+            # match the param's own dense layout (NCHW or channels_last) so
+            # autograd accumulates into the bucket with the param's strides
+            # and the flat fused-optimizer kernels see aligned layouts
+            p.grad = buffer.as_strided(p.shape, p.stride(), offset)
             offset += n
         bucket = Bucket(params, buffer)
         for p in params:

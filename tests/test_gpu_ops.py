@@ -141,29 +141,36 @@ class TestGNS:
 
 class TestTrainingParity:
     def test_resnet_step_matches_eager(self):
-        """One full fused-optimizer training step == torch.optim step."""
+        """Fused-optimizer step == torch.optim step on REAL model
+        gradients.  One model produces the gradients (MIOpen's split-K
+        weight-gradient kernels accumulate non-deterministically, so a
+        second backward would differ on its own); both optimizers then
+        step identical parameter copies with those shared gradients."""
         from shockwave_amd.models import resnet18_cifar
         from shockwave_amd.ops.optim import FusedSGD
 
         torch.manual_seed(7)
-        m1 = resnet18_cifar().to(DEV)
-        m2 = resnet18_cifar().to(DEV)
-        m2.load_state_dict(m1.state_dict())
-        o1 = FusedSGD(m1.parameters(), lr=0.1, momentum=0.9, weight_decay=5e-4)
-        o2 = torch.optim.SGD(
-            m2.parameters(), lr=0.1, momentum=0.9, weight_decay=5e-4
-        )
+        model = resnet18_cifar().to(DEV)
         x = torch.randn(8, 3, 32, 32, device=DEV)
         y = torch.randint(0, 10, (8,), device=DEV)
-        for _ in range(3):
-            for m, o in ((m1, o1), (m2, o2)):
-                o.zero_grad(set_to_none=False)
-                torch.nn.functional.cross_entropy(m(x), y).backward()
-                o.step()
+        torch.nn.functional.cross_entropy(model(x), y).backward()
         torch.cuda.synchronize()
-        for (n1, p1), (n2, p2) in zip(
-            m1.named_parameters(), m2.named_parameters()
-        ):
-            # 3 training steps amplify fp32 fmaf-vs-eager rounding through
-            # BN statistics; tolerance reflects that, not kernel error
-            torch.testing.assert_close(p1, p2, rtol=5e-2, atol=1e-4), n1
+
+        names = [n for n, p in model.named_parameters()]
+        grads = {n: p.grad.clone() for n, p in model.named_parameters()}
+
+        p1 = {n: p.detach().clone().requires_grad_() for n, p in model.named_parameters()}
+        p2 = {n: p.detach().clone().requires_grad_() for n, p in model.named_parameters()}
+        o1 = FusedSGD(p1.values(), lr=0.1, momentum=0.9, weight_decay=5e-4)
+        o2 = torch.optim.SGD(p2.values(), lr=0.1, momentum=0.9, weight_decay=5e-4)
+        for _ in range(3):
+            for n in names:
+                p1[n].grad = grads[n].clone()
+                p2[n].grad = grads[n].clone()
+            o1.step()
+            o2.step()
+        torch.cuda.synchronize()
+        for n in names:
+            torch.testing.assert_close(
+                p1[n], p2[n], rtol=1e-4, atol=1e-6, msg=lambda m: f"{n}: {m}"
+            )
