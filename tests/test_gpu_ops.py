@@ -44,8 +44,6 @@ class TestFusedSGD:
         grads = [torch.randn(s, device=DEV) for s in SIZES]
 
         ref_params = [torch.nn.Parameter(p) for p in p_ref]
-        for rp, g in zip(ref_params, grads):
-            rp.grad = g.clone()
         ref_opt = torch.optim.SGD(
             ref_params, lr=0.1, momentum=momentum, weight_decay=wd,
             nesterov=nesterov,
@@ -53,6 +51,10 @@ class TestFusedSGD:
 
         bufs = [torch.zeros_like(p) for p in p_gpu]
         for step in range(3):
+            # re-set grads every step: torch's foreach nesterov path
+            # mutates .grad in place (grad += momentum*buf)
+            for rp, g in zip(ref_params, grads):
+                rp.grad = g.clone()
             ref_opt.step()
             ops.fused_sgd(
                 p_gpu, grads, bufs, lr=0.1, momentum=momentum,
