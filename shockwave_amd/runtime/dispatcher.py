@@ -1,0 +1,232 @@
+"""Per-worker job dispatcher.
+
+Rebuild of the reference dispatcher (runtime/rpc/dispatcher.py:1-634):
+
+* GPU-slot queue: one slot per local GPU, checked out per job
+  (:64-69, :521),
+* builds the shell command — job command + ``--local_rank`` +
+  ``<num_steps_arg> N`` + ``--checkpoint_dir`` + ``--enable_gavel_iterator``
+  (:179-206) — and runs it from the mode-specific run dir (:352-358),
+* exports the iterator env contract (GAVEL_JOB_ID/WORKER_ID/ROUND_ID/
+  SCHED_ADDR/SCHED_PORT) and pins the GPU via HIP/ROCR_VISIBLE_DEVICES
+  (the reference pins CUDA_VISIBLE_DEVICES; :385-399),
+* scrapes the iterator round log for (steps, duration) (:208-237),
+* kills job processes by tracked process group (:239-295; the reference
+  greps command lines for pids — tracking the Popen handle is safer),
+* reports Done to the scheduler with the full iterator log (:611).
+
+CUDA MPS has no ROCm equivalent and is not needed: packed jobs simply
+share the GPU, and HIP's hardware scheduler time-slices them (SURVEY §2.4
+row 10).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import re
+import signal
+import subprocess
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from typing import Dict, List, Optional
+
+from .set_queue import SetQueue
+
+logger = logging.getLogger("shockwave_amd.dispatcher")
+
+LOG_LINE_RE = re.compile(
+    r"^\[(?P<ts>[^\]]+)\] \[(?P<event>[^\]]+)\] \[(?P<status>[^\]]+)\] ?(?P<msg>.*)$"
+)
+
+
+class Dispatcher:
+    def __init__(
+        self,
+        round_duration: float,
+        gpu_ids: List[int],
+        worker_rpc_client,
+        sched_addr: str,
+        sched_port: int,
+        run_dir: str,
+        data_dir: Optional[str],
+        checkpoint_dir: str,
+        static_run_dir: Optional[str] = None,
+        accordion_run_dir: Optional[str] = None,
+        gns_run_dir: Optional[str] = None,
+    ):
+        self._round_duration = round_duration
+        self._gpu_ids = gpu_ids
+        self._gpu_queue = SetQueue()
+        for gpu_id in gpu_ids:
+            self._gpu_queue.put(gpu_id)
+        self._worker_rpc_client = worker_rpc_client
+        self._sched_addr = sched_addr
+        self._sched_port = sched_port
+        self._run_dir = run_dir
+        self._data_dir = data_dir
+        self._checkpoint_dir = checkpoint_dir
+        self._run_dirs = {
+            "static": static_run_dir or run_dir,
+            "accordion": accordion_run_dir or run_dir,
+            "gns": gns_run_dir or run_dir,
+        }
+        self._lock = threading.Lock()
+        self._procs: Dict[int, subprocess.Popen] = {}  # job_id -> proc
+        self._killed = set()
+        self._pool = ThreadPoolExecutor(max_workers=max(8, 2 * len(gpu_ids)))
+
+    # -- command construction ----------------------------------------------
+
+    def _construct_command(self, job, gpu_id: int, worker_id: int) -> str:
+        command = job["command"]
+        if job.get("needs_data_dir") and self._data_dir and "%s" in command:
+            n = command.count("%s")
+            command = command % tuple([self._data_dir] * n)
+        command = f"{command} --local_rank {gpu_id}"
+        command = f"{command} {job['num_steps_arg']} {job['num_steps']}"
+        ckpt = os.path.join(self._checkpoint_dir, f"job_id={job['job_id']}")
+        command = f"{command} --checkpoint_dir {ckpt}"
+        command = f"{command} --enable_gavel_iterator"
+        mode = job.get("mode")
+        if mode:
+            command = f"{command} --mode {mode}"
+        return command
+
+    def _job_working_dir(self, job) -> str:
+        base = self._run_dirs.get(job.get("mode", "static"), self._run_dir)
+        return os.path.join(base, job["working_directory"])
+
+    # -- log scraping --------------------------------------------------------
+
+    def _get_steps_and_execution_time(self, job_id, worker_id, round_id):
+        log_file = os.path.join(
+            self._checkpoint_dir,
+            f"job_id={job_id}",
+            ".gavel",
+            f"round={round_id}",
+            f"worker={worker_id}.log",
+        )
+        steps, duration, lines = 0, 0.0, []
+        try:
+            with open(log_file) as f:
+                for line in f:
+                    lines.append(line.rstrip("\n"))
+                    m = LOG_LINE_RE.match(line)
+                    if not m:
+                        continue
+                    if m.group("event") == "PROGRESS":
+                        if m.group("status") == "STEPS":
+                            steps = int(float(m.group("msg")))
+                        elif m.group("status") == "DURATION":
+                            duration = float(m.group("msg"))
+        except FileNotFoundError:
+            logger.warning("no iterator log at %s", log_file)
+        return steps, duration, "\n".join(lines)
+
+    # -- launch / kill -------------------------------------------------------
+
+    def launch_job(self, job, command, worker_id, round_id, gpu_id):
+        job_id = job["job_id"]
+        env = dict(os.environ)
+        env.update(
+            {
+                "GAVEL_JOB_ID": str(job_id),
+                "GAVEL_WORKER_ID": str(worker_id),
+                "GAVEL_ROUND_ID": str(round_id),
+                "GAVEL_SCHED_ADDR": self._sched_addr,
+                "GAVEL_SCHED_PORT": str(self._sched_port),
+                "HIP_VISIBLE_DEVICES": str(gpu_id),
+                "ROCR_VISIBLE_DEVICES": str(gpu_id),
+                "CUDA_VISIBLE_DEVICES": str(gpu_id),
+                "SWQ_MODE": job.get("mode", "static"),
+            }
+        )
+        cwd = self._job_working_dir(job)
+        logger.info(
+            "[worker %s round %s] launching job %s on gpu %s: %s (cwd %s)",
+            worker_id, round_id, job_id, gpu_id, command, cwd,
+        )
+        proc = subprocess.Popen(
+            command,
+            shell=True,
+            cwd=cwd,
+            env=env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT,
+            start_new_session=True,  # own process group for clean kills
+        )
+        with self._lock:
+            self._procs[job_id] = proc
+        stdout, _ = proc.communicate()
+        with self._lock:
+            self._procs.pop(job_id, None)
+            was_killed = job_id in self._killed
+            self._killed.discard(job_id)
+        if proc.returncode != 0 and not was_killed:
+            logger.error(
+                "job %s exited rc=%s; output tail:\n%s",
+                job_id, proc.returncode,
+                b"\n".join(stdout.splitlines()[-20:]).decode(errors="replace"),
+            )
+        return proc.returncode, stdout
+
+    def _dispatch_job_helper(self, job, worker_id, round_id):
+        gpu_id = self._gpu_queue.get()
+        try:
+            command = self._construct_command(job, gpu_id, worker_id)
+            self.launch_job(job, command, worker_id, round_id, gpu_id)
+            steps, duration, log = self._get_steps_and_execution_time(
+                job["job_id"], worker_id, round_id
+            )
+        finally:
+            self._gpu_queue.put(gpu_id)
+        self._worker_rpc_client.notify_scheduler(
+            worker_id, [(job["job_id"], steps, duration, log)]
+        )
+
+    def dispatch_jobs(self, job_descriptions, worker_id, round_id):
+        for job in job_descriptions:
+            self._pool.submit(
+                self._safe_dispatch, job, worker_id, round_id
+            )
+
+    def _safe_dispatch(self, job, worker_id, round_id):
+        try:
+            self._dispatch_job_helper(job, worker_id, round_id)
+        except Exception:
+            logger.exception("dispatch of job %s failed", job.get("job_id"))
+            try:
+                self._worker_rpc_client.notify_scheduler(
+                    worker_id, [(job["job_id"], 0, 0.0, "")]
+                )
+            except Exception:
+                logger.exception("failed to notify scheduler of failure")
+
+    def kill_job(self, job_id):
+        with self._lock:
+            proc = self._procs.get(job_id)
+            if proc is not None:
+                self._killed.add(job_id)
+        if proc is None:
+            logger.info("kill_job(%s): no running process", job_id)
+            return
+        logger.info("killing job %s (pid %s)", job_id, proc.pid)
+        try:
+            os.killpg(os.getpgid(proc.pid), signal.SIGTERM)
+            try:
+                proc.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                os.killpg(os.getpgid(proc.pid), signal.SIGKILL)
+        except ProcessLookupError:
+            pass
+
+    def reset(self):
+        with self._lock:
+            job_ids = list(self._procs.keys())
+        for job_id in job_ids:
+            self.kill_job(job_id)
+
+    def shutdown(self):
+        self.reset()
+        self._pool.shutdown(wait=False)
