@@ -1,0 +1,172 @@
+"""Generic lease-preemptible training loop.
+
+One implementation of the epoch/lease/checkpoint/adaptation state machine
+all families share (the reference repeats it in 21 mains across three
+workload trees).  A family provides a ``WorkloadSpec``; the loop handles:
+
+* device + RCCL (or gloo) distributed bring-up,
+* checkpoint restore/save (model, optimizer, epoch, steps, adaptation
+  state) through the LeaseIterator hooks,
+* the Accordion / GNS shared adaptation library and
+  ``update_resource_requirement`` reporting,
+* per-step gradient sync + fused-optimizer stepping + throughput lines.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import Callable, Optional
+
+import torch
+
+from ..adapt import AccordionDetector, GNSEstimator, hardcoded_critical_regime
+from ..core import datasets as ds_tables
+from . import common
+
+
+@dataclasses.dataclass
+class WorkloadSpec:
+    family: str                      # "ResNet-18", "LM", ...
+    build_model: Callable            # (args, device) -> nn.Module
+    build_loader: Callable           # (args) -> iterable of batches
+    build_optimizer: Callable        # (args, params) -> optimizer
+    step: Callable                   # (model, batch, device, state) -> loss
+    supports_accordion: bool = True
+    supports_gns: bool = True
+    synthetic_data: bool = True
+    extra_state: Callable = None     # (state) -> dict to checkpoint
+    restore_state: Callable = None   # (state, dict) -> None
+
+
+def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
+        max_steps_override: Optional[int] = None):
+    import os
+
+    mode = args.mode or mode or os.environ.get("SWQ_MODE", "static")
+    device = common.init_device_and_distributed(args)
+
+    loader = spec.build_loader(args)
+    model = spec.build_model(args, device)
+    model_train = common.wrap_distributed(model, args)
+    optimizer = spec.build_optimizer(args, model_train.parameters())
+
+    trainloader, lease_it = common.make_lease_iterator(
+        loader, args, synthetic_data=spec.synthetic_data, client=client
+    )
+
+    state = {
+        "args": args,
+        "device": device,
+        "model": model,
+        "epoch": 0,
+        "cumulative_steps": 0,
+        "original_bs": getattr(args, "batch_size", 0),
+    }
+
+    accordion = gns = None
+    if mode == "accordion" and spec.supports_accordion:
+        accordion = AccordionDetector(model)
+    elif mode == "gns" and spec.supports_gns:
+        window = args.world_size if args.world_size > 1 else 2
+        gns = GNSEstimator(model, getattr(args, "batch_size", 1), window=window)
+
+    ckpt = lease_it.load_checkpoint() if lease_it is not None else None
+    if ckpt:
+        model.load_state_dict(ckpt["model"])
+        try:
+            optimizer.load_state_dict(ckpt["optimizer"])
+        except (ValueError, KeyError):
+            pass
+        state["epoch"] = ckpt.get("epoch", 0)
+        state["cumulative_steps"] = ckpt.get("cumulative_steps", 0)
+        state["original_bs"] = ckpt.get("original_bs", state["original_bs"])
+        if accordion is not None and "accordion" in ckpt:
+            accordion.load_state_dict(ckpt["accordion"])
+        if gns is not None and "gns" in ckpt:
+            gns.load_state_dict(ckpt["gns"])
+        if spec.restore_state and "extra" in ckpt:
+            spec.restore_state(state, ckpt["extra"])
+
+    steps_per_epoch = max(1, len(loader))
+    target_steps = (
+        max_steps_override or args.num_steps or steps_per_epoch * 5
+    )
+    reporter = common.ThroughputReporter(
+        args.throughput_estimation_interval, args.rank
+    )
+
+    def save_checkpoint():
+        if args.rank != 0 or lease_it is None:
+            return
+        out = {
+            "model": model.state_dict(),
+            "optimizer": optimizer.state_dict(),
+            "epoch": state["epoch"],
+            "cumulative_steps": state["cumulative_steps"],
+            "original_bs": state["original_bs"],
+        }
+        if accordion is not None:
+            out["accordion"] = accordion.state_dict()
+        if gns is not None:
+            out["gns"] = gns.state_dict()
+        if spec.extra_state:
+            out["extra"] = spec.extra_state(state)
+        lease_it.save_checkpoint(out)
+
+    def maybe_request_rescale(epoch):
+        """Epoch-boundary batch-size decisions (shared across families)."""
+        if lease_it is None:
+            return False
+        bs = getattr(args, "batch_size", 0)
+        max_bs = ds_tables.max_batch_size(spec.family, bs)
+        if accordion is not None:
+            accordion.on_epoch(epoch)
+            in_cr = hardcoded_critical_regime(
+                spec.family, state["original_bs"], epoch + 1
+            )
+            if not in_cr and bs == state["original_bs"] and bs != max_bs:
+                lease_it.update_resource_requirement(True, False)
+                return True
+            if in_cr and bs != state["original_bs"]:
+                lease_it.update_resource_requirement(False, True)
+                return True
+        if gns is not None:
+            gns.on_epoch(epoch)
+            if gns.should_double(epoch) and bs < max_bs:
+                lease_it.update_resource_requirement(True, False)
+                return True
+        return False
+
+    model_train.train()
+    done = False
+    while not done and state["cumulative_steps"] < target_steps:
+        hit_target = False
+        for batch in trainloader:
+            common.zero_grads(model_train)
+            loss = spec.step(model_train, batch, device, state)
+            if loss is not None and loss.requires_grad:
+                loss.backward()
+            common.finish_sync(model_train)
+            if accordion is not None:
+                accordion.on_step()
+            if gns is not None:
+                gns.on_step()
+            optimizer.step()
+            state["cumulative_steps"] += 1
+            reporter.step()
+            if state["cumulative_steps"] >= target_steps:
+                hit_target = True
+                break
+        if hit_target or (lease_it is not None and lease_it.done):
+            break
+        if maybe_request_rescale(state["epoch"]):
+            done = True
+        state["epoch"] += 1
+
+    if lease_it is not None:
+        save_checkpoint()
+        if state["cumulative_steps"] >= target_steps and not lease_it.done:
+            lease_it.complete()
+    if torch.distributed.is_initialized():
+        torch.distributed.destroy_process_group()
+    return state["cumulative_steps"]

@@ -1,0 +1,91 @@
+"""End-to-end physical-mode test: head + worker + real training subprocesses
+over the live gRPC control plane, on CPU (BASELINE config 2 shape, scaled
+down)."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import pytest
+
+from shockwave_amd.core.job import Job
+from shockwave_amd.core import trace as trace_mod
+
+REPO = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def tiny_job(steps=6, bs=16):
+    return Job(
+        job_id=None,
+        job_type=f"ResNet-18 (batch size {bs})",
+        command=f"python3 main.py --batch_size {bs}",
+        working_directory="image_classification/cifar10",
+        num_steps_arg="--num_steps",
+        total_steps=steps,
+        duration=600,
+        scale_factor=1,
+        mode="static",
+    )
+
+
+@pytest.mark.slow
+class TestPhysicalEndToEnd:
+    def test_single_job_completes_through_rpc(self, tmp_path, throughputs):
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        port = free_port()
+        worker_port = free_port()
+        jobs = [tiny_job(steps=6)]
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+
+        sched = PhysicalScheduler(
+            get_policy("max_min_fairness"),
+            port=port,
+            expected_num_workers=1,
+            throughputs=throughputs,
+            time_per_iteration=30,
+            profiles=profiles,
+            worker_type="mi355x",
+        )
+        try:
+            worker = Worker(
+                worker_type="mi355x",
+                sched_addr="127.0.0.1",
+                sched_port=port,
+                worker_port=worker_port,
+                num_gpus=1,
+                ip_addr="127.0.0.1",
+                run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                accordion_run_dir=os.path.join(REPO, "workloads", "accordion"),
+                gns_run_dir=os.path.join(REPO, "workloads", "gns"),
+                checkpoint_dir=str(tmp_path),
+            )
+            assert worker.round_duration == 30
+            sched.add_job(jobs[0])
+
+            deadline = time.time() + 240
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(2)
+            completions = sched.get_job_completion_times()
+            assert len(completions) == 1, "job did not complete"
+            (jct,) = completions.values()
+            assert 0 < jct < 240
+            # the job ran over the real dispatcher: its checkpoint exists
+            assert os.path.exists(
+                os.path.join(str(tmp_path), "job_id=0", "model.chkpt")
+            )
+        finally:
+            sched.shutdown()

@@ -1,0 +1,18 @@
+"""Trace-compatible entry shim -> shockwave_amd.workloads.families.cifar10_main
+(mode: static).  Keeps the reference's run-dir/CLI layout so its traces
+dispatch unchanged."""
+import os
+import sys
+
+_d = os.path.dirname(os.path.abspath(__file__))
+while not os.path.isdir(os.path.join(_d, "shockwave_amd")):
+    parent = os.path.dirname(_d)
+    if parent == _d:
+        raise RuntimeError("repo root not found")
+    _d = parent
+sys.path.insert(0, _d)
+
+from shockwave_amd.workloads.families import cifar10_main
+
+if __name__ == "__main__":
+    sys.exit(0 if cifar10_main(mode="static") is not None else 1)
