@@ -152,6 +152,12 @@ def serve_worker(port: int, callbacks: Dict[str, Callable]):
         callbacks["Shutdown"]()
         return {}
 
+    def fetch_checkpoint(req):
+        fn = callbacks.get("FetchCheckpoint")
+        if fn is None:
+            return {"found": False, "data": b"", "total": 0}
+        return fn(req["job_id"], req["offset"], req["length"])
+
     return make_server(
         port,
         {
@@ -160,6 +166,7 @@ def serve_worker(port: int, callbacks: Dict[str, Callable]):
                 "KillJob": kill_job,
                 "Reset": reset,
                 "Shutdown": shutdown,
+                "FetchCheckpoint": fetch_checkpoint,
             }
         },
     )

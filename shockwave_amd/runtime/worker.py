@@ -46,6 +46,7 @@ class Worker:
         self.worker_type = worker_type
         self.num_gpus = num_gpus
 
+        self._checkpoint_dir = checkpoint_dir
         self._server = serve_worker(
             worker_port,
             {
@@ -53,6 +54,7 @@ class Worker:
                 "KillJob": self._kill_job_callback,
                 "Reset": self._reset_callback,
                 "Shutdown": self._shutdown_callback,
+                "FetchCheckpoint": self._fetch_checkpoint_callback,
             },
         )
 
@@ -110,6 +112,23 @@ class Worker:
     def _shutdown_callback(self):
         self._dispatcher.shutdown()
         self._done.set()
+
+    def _fetch_checkpoint_callback(self, job_id, offset, length):
+        """Serve checkpoint bytes for cross-node migration
+        (shockwave_amd/parallel/ckpt_stream.py)."""
+        from ..parallel.ckpt_stream import CheckpointStore
+
+        store = CheckpointStore(
+            os.path.join(self._checkpoint_dir, f"job_id={job_id}")
+        )
+        data = store.read_bytes()
+        if data is None:
+            return {"found": False, "data": b"", "total": 0}
+        return {
+            "found": True,
+            "data": data[offset : offset + length],
+            "total": len(data),
+        }
 
     def join(self):
         self._done.wait()

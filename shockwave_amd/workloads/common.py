@@ -89,21 +89,23 @@ def checkpoint_path(checkpoint_dir: str) -> str:
 
 
 def default_load_checkpoint(path):
+    from ..parallel.ckpt_stream import CheckpointStore
+
+    store = CheckpointStore(os.path.dirname(path))
+
     def load():
-        if path and os.path.exists(path):
-            return torch.load(path, map_location="cpu", weights_only=False)
-        return None
+        return store.load()
 
     return load
 
 
 def default_save_checkpoint(path):
+    from ..parallel.ckpt_stream import CheckpointStore
+
+    store = CheckpointStore(os.path.dirname(path))
+
     def save(state):
-        if path:
-            os.makedirs(os.path.dirname(path), exist_ok=True)
-            tmp = path + ".tmp"
-            torch.save(state, tmp)
-            os.replace(tmp, path)
+        store.save(state)
 
     return save
 
