@@ -12,8 +12,9 @@ steps/s = sf * isolated * eff(sf), with efficiencies reflecting 7-link xGMI
 all-reduce overlap (far better than the reference's PCIe/IB V100 numbers,
 where DDP cost ~30-50% per doubling).
 
-Pairwise (packing) entries use a utilization contention model: a job packed
-with a partner of utilization u runs at iso/(1 + u/100).
+Pairwise (packing) entries use a utilization contention model: two packed
+jobs each run at iso/(1 + (u_a/100)*(u_b/100)) — contention scales with
+the product of their GPU utilizations.
 """
 
 import json
@@ -105,8 +106,11 @@ def main(out_path="traces/mi355x_throughputs.json"):
                 other_bs = int(jt2[jt2.rfind(" ") + 1 : -1])
                 u_mine = datasets.util_pct(my_model, my_bs)
                 u_theirs = datasets.util_pct(other_model, other_bs)
-                mine = entries[(jt, 1)] / (1.0 + u_theirs / 100.0)
-                theirs = entries[(jt2, 1)] / (1.0 + u_mine / 100.0)
+                # contention grows with the PRODUCT of utilizations, so a
+                # job's colocation fingerprint depends on its own intensity
+                contention = (u_mine / 100.0) * (u_theirs / 100.0)
+                mine = entries[(jt, 1)] / (1.0 + contention)
+                theirs = entries[(jt2, 1)] / (1.0 + contention)
                 e[format_job_type_key((jt2, 1))] = [mine, theirs]
         raw[WORKER_TYPE][format_job_type_key((jt, sf))] = e
 
