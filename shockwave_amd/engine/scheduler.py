@@ -1098,23 +1098,29 @@ class RoundScheduler:
     # ------------------------------------------------------------------
 
     def simulate(self, cluster_spec, arrival_times, jobs,
-                 num_gpus_per_server=None, debug=False):
-        queued_jobs = list(zip(arrival_times, jobs))
-        remaining_jobs = len(jobs)
+                 num_gpus_per_server=None, debug=False,
+                 checkpoint_threshold=None, checkpoint_file=None,
+                 _resume_state=None):
+        if _resume_state is None:
+            queued_jobs = list(zip(arrival_times, jobs))
+            remaining_jobs = len(jobs)
+            current_round = 0
+            for worker_type in sorted(cluster_spec):
+                num_gpus = 1
+                if num_gpus_per_server:
+                    num_gpus = num_gpus_per_server[worker_type]
+                for _ in range(cluster_spec[worker_type] // num_gpus):
+                    self.register_worker(worker_type, num_gpus=num_gpus)
+            if queued_jobs:
+                self._current_timestamp = queued_jobs[0][0]
+        else:
+            queued_jobs = _resume_state["queued_jobs"]
+            remaining_jobs = _resume_state["remaining_jobs"]
+            current_round = _resume_state["current_round"]
         running_jobs = []  # heap of (-finish_time, job_id, worker_ids, steps)
         current_round_start_time = 0
         current_round_end_time = None
-
-        for worker_type in sorted(cluster_spec):
-            num_gpus = 1
-            if num_gpus_per_server:
-                num_gpus = num_gpus_per_server[worker_type]
-            for _ in range(cluster_spec[worker_type] // num_gpus):
-                self.register_worker(worker_type, num_gpus=num_gpus)
-
-        if queued_jobs:
-            self._current_timestamp = queued_jobs[0][0]
-        current_round = 0
+        checkpoint_saved = False
 
         while True:
             if remaining_jobs == 0:
@@ -1205,6 +1211,20 @@ class RoundScheduler:
                 logger.warning("simulation complete: no jobs left")
                 break
 
+            # mid-trace checkpoint for long sweeps (reference
+            # _save_checkpoint/_load_checkpoint :1518-1594): taken at a round
+            # boundary where no micro-tasks are in flight
+            if (
+                checkpoint_file is not None
+                and checkpoint_threshold is not None
+                and not checkpoint_saved
+                and self._job_id_counter >= checkpoint_threshold
+            ):
+                self.save_simulation_checkpoint(
+                    checkpoint_file, queued_jobs, remaining_jobs, current_round
+                )
+                checkpoint_saved = True
+
             # schedule the round
             scheduled_jobs = self._schedule_jobs_on_workers()
             if not scheduled_jobs and self._jobs and not queued_jobs:
@@ -1253,6 +1273,38 @@ class RoundScheduler:
             self._current_timestamp / 3600.0,
         )
         return self._current_timestamp
+
+    # ------------------------------------------------------------------
+    # Simulation checkpointing (reference :1518-1594)
+    # ------------------------------------------------------------------
+
+    def save_simulation_checkpoint(self, path, queued_jobs, remaining_jobs,
+                                   current_round):
+        import pickle
+
+        with open(path, "wb") as f:
+            pickle.dump(
+                {
+                    "scheduler": self,
+                    "queued_jobs": queued_jobs,
+                    "remaining_jobs": remaining_jobs,
+                    "current_round": current_round,
+                },
+                f,
+            )
+        logger.info("simulation checkpoint -> %s (round %d)", path,
+                    current_round)
+
+    @staticmethod
+    def resume_simulation(path):
+        """Returns (scheduler, resume_state); continue with
+        ``sched.simulate(cluster_spec, None, None, _resume_state=state)``."""
+        import pickle
+
+        with open(path, "rb") as f:
+            saved = pickle.load(f)
+        sched = saved.pop("scheduler")
+        return sched, saved
 
     # ------------------------------------------------------------------
     # Metrics (reference :2779-3107)

@@ -108,3 +108,38 @@ class TestDynamicAdaptation:
         # (job completed, so check trace of bs via original records)
         assert job.batch_size > 16
         assert sched.is_done()
+
+
+class TestSimulationCheckpoint:
+    def test_checkpoint_and_resume(self, tmp_path, throughputs):
+        """A sweep interrupted mid-trace resumes and finishes all jobs."""
+        import os
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [
+            trace_mod.build_job_profile(j, throughputs) for j in jobs
+        ]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+        ckpt = str(tmp_path / "sim.ckpt")
+
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        full_makespan = sched.simulate({"mi355x": 2}, arrivals, jobs,
+                                       checkpoint_threshold=5,
+                                       checkpoint_file=ckpt)
+        assert os.path.exists(ckpt)
+        assert len(sched.get_job_completion_times()) == 8
+
+        resumed, state = RoundScheduler.resume_simulation(ckpt)
+        makespan = resumed.simulate({"mi355x": 2}, None, None,
+                                    _resume_state=state)
+        assert len(resumed.get_job_completion_times()) == 8
+        # resumed run finishes at the same simulated makespan
+        assert abs(makespan - full_makespan) / full_makespan < 0.05
