@@ -47,18 +47,44 @@ STEPS_ARG = {
 
 
 def measure_one(family, bs, steps, warmup):
+    """steps/s measured from the lease iterator's own PROGRESS log, which
+    times the training loop only (model construction, synthetic-data
+    generation and MIOpen algorithm search are excluded: the warmup call
+    absorbs them, and the timed call's iterator clock starts at the loop)."""
+    import shutil
+    import tempfile
+
+    from shockwave_amd.runtime.dispatcher import LOG_LINE_RE
     from shockwave_amd.workloads import families as fam_mod
 
     fn_name, make_args = FAMILY_ARGS[family]
     fn = getattr(fam_mod, fn_name)
-    argv = make_args(bs) + [STEPS_ARG[family], str(warmup)]
-    # warmup run (algo selection, allocator) — untimed
-    fn(argv)
-    t0 = time.time()
-    argv = make_args(bs) + [STEPS_ARG[family], str(steps)]
-    done = fn(argv)
-    elapsed = time.time() - t0
-    return done / elapsed
+    # warmup run (MIOpen find, allocator, L3) — untimed
+    fn(make_args(bs) + [STEPS_ARG[family], str(warmup)])
+
+    ckpt_dir = tempfile.mkdtemp(prefix="swq_measure_")
+    try:
+        argv = make_args(bs) + [
+            STEPS_ARG[family], str(steps),
+            "--checkpoint_dir", ckpt_dir, "--enable_gavel_iterator",
+        ]
+        from shockwave_amd.runtime.lease_iterator import NullLeaseClient
+
+        done = fn(argv, client=NullLeaseClient())
+        log_path = os.path.join(ckpt_dir, ".gavel", "round=0", "worker=0.log")
+        steps_logged, duration = done, None
+        with open(log_path) as f:
+            for line in f:
+                m = LOG_LINE_RE.match(line)
+                if m and m.group("event") == "PROGRESS":
+                    if m.group("status") == "STEPS":
+                        steps_logged = int(float(m.group("msg")))
+                    elif m.group("status") == "DURATION":
+                        duration = float(m.group("msg"))
+        assert duration and duration > 0, "no duration in iterator log"
+        return steps_logged / duration
+    finally:
+        shutil.rmtree(ckpt_dir, ignore_errors=True)
 
 
 def main():

@@ -198,6 +198,12 @@ class LeaseIterator:
     def duration(self):
         return self._duration
 
+    def write_progress(self):
+        """Flush a PROGRESS STEPS/DURATION pair to the round log (the
+        dispatcher and profiling harness scrape these)."""
+        self._write_info()
+        self._file_handler.flush()
+
     def complete(self, timeout=False):
         self._done = True
         if not self._write_on_close:

@@ -167,6 +167,7 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
         save_checkpoint()
         if state["cumulative_steps"] >= target_steps and not lease_it.done:
             lease_it.complete()
+        lease_it.write_progress()
     if torch.distributed.is_initialized():
         torch.distributed.destroy_process_group()
     return state["cumulative_steps"]
