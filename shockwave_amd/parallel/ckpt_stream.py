@@ -38,8 +38,14 @@ class CheckpointStore:
         shared_dir: Optional[str] = None,
     ):
         self.dir = job_checkpoint_dir
+        # key the RAM tier by the full durable path so distinct checkpoint
+        # dirs (and re-created temp dirs) never share shm state
+        import hashlib
+
+        abspath = os.path.abspath(job_checkpoint_dir)
+        digest = hashlib.sha1(abspath.encode()).hexdigest()[:16]
         self.shm_dir = os.path.join(
-            shm_root, os.path.basename(os.path.normpath(job_checkpoint_dir))
+            shm_root, f"{os.path.basename(os.path.normpath(abspath))}-{digest}"
         )
         self.shared_dir = shared_dir
         os.makedirs(self.dir, exist_ok=True)
@@ -73,19 +79,32 @@ class CheckpointStore:
             os.replace(tmp, shared_path)
         return dir_path
 
+    def _read_order(self):
+        """The shm entry is a CACHE of the durable file: use it only when
+        the durable copy exists and is not newer (a deleted/recreated job
+        dir must never resurrect a stale RAM-tier checkpoint)."""
+        shm_path, dir_path, shared_path = self._paths()
+        order = []
+        if os.path.exists(dir_path):
+            if (
+                os.path.exists(shm_path)
+                and os.path.getmtime(shm_path) >= os.path.getmtime(dir_path)
+            ):
+                order.append(shm_path)
+            order.append(dir_path)
+        if shared_path and os.path.exists(shared_path):
+            order.append(shared_path)
+        return order
+
     def load(self) -> Optional[dict]:
-        for path in self._paths():
-            if path and os.path.exists(path):
-                return torch.load(
-                    path, map_location="cpu", weights_only=False
-                )
+        for path in self._read_order():
+            return torch.load(path, map_location="cpu", weights_only=False)
         return None
 
     def read_bytes(self) -> Optional[bytes]:
-        for path in self._paths():
-            if path and os.path.exists(path):
-                with open(path, "rb") as f:
-                    return f.read()
+        for path in self._read_order():
+            with open(path, "rb") as f:
+                return f.read()
         return None
 
     def write_bytes(self, data: bytes) -> None:
