@@ -15,12 +15,16 @@ then the unbiased estimators (OpenAI GNS):
     |G|^2 = (B_big |G_big|^2 - B_small |G_small|^2) / (B_big - B_small)
     S     = (|G_small|^2 - |G_big|^2) / (1/B_small - 1/B_big)
     GNS   = S / |G|^2
+
+Per-step state (the EMA of |G|^2 and S) stays in DEVICE tensors so the
+training loop never synchronizes; the host reads one scalar per epoch
+(``on_epoch``), keeping the estimator hipGraph- and overlap-friendly.
 """
 
 from __future__ import annotations
 
 from collections import deque
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
@@ -35,46 +39,55 @@ class GNSEstimator:
         self.params = [p for p in model.parameters() if p.requires_grad]
         self._grads: deque = deque(maxlen=self.window)
         self.ema = ema
-        self._g2_avg: Optional[float] = None
-        self._s_avg: Optional[float] = None
+        self._device = self.params[0].device
+        # running EMA of (|G|^2 estimate, S estimate) — device-resident
+        self._avg = torch.zeros(2, device=self._device)
+        self._have_avg = False
         self.gns_by_epoch = {}
 
     def _flat_grad(self) -> torch.Tensor:
-        return torch.cat(
+        flat = torch.cat(
             [
                 p.grad.detach().reshape(-1).float()
                 for p in self.params
                 if p.grad is not None
             ]
         )
+        return flat
 
-    def on_step(self) -> Optional[float]:
-        """Push the current gradient; return the running GNS estimate once
-        the window is full."""
-        self._grads.append(self._flat_grad())
+    def on_step(self) -> None:
+        """Push the current gradient and fold the unbiased estimates into
+        the device-side EMA.  No host synchronization."""
+        # the window must hold COPIES: grad buffers are reused every step
+        slot = self._flat_grad().clone()
+        self._grads.append(slot)
         if len(self._grads) < self.window:
-            return None
+            return
         big_sq, small_sq = ops.gns_window_stats(list(self._grads))
-        big_sq = float(big_sq)
-        small_sq = float(small_sq)
-        b_small = self.batch_size
-        b_big = self.batch_size * self.window
+        b_small = float(self.batch_size)
+        b_big = float(self.batch_size * self.window)
         g2 = (b_big * big_sq - b_small * small_sq) / (b_big - b_small)
         s = (small_sq - big_sq) / (1.0 / b_small - 1.0 / b_big)
-        if self._g2_avg is None:
-            self._g2_avg, self._s_avg = g2, s
+        cur = torch.stack([g2, s]) if torch.is_tensor(g2) else torch.tensor(
+            [g2, s], device=self._device
+        )
+        if not self._have_avg:
+            self._avg.copy_(cur)
+            self._have_avg = True
         else:
-            self._g2_avg = self.ema * self._g2_avg + (1 - self.ema) * g2
-            self._s_avg = self.ema * self._s_avg + (1 - self.ema) * s
-        if self._g2_avg <= 0:
+            self._avg.mul_(self.ema).add_(cur, alpha=1 - self.ema)
+
+    def current_gns(self) -> Optional[float]:
+        """Host-side read (synchronizes); call at epoch boundaries only."""
+        if not self._have_avg:
             return None
-        return self._s_avg / self._g2_avg
+        g2, s = self._avg.tolist()
+        if g2 <= 0:
+            return None
+        return s / g2
 
     def on_epoch(self, epoch: int) -> None:
-        gns = None
-        if self._g2_avg and self._g2_avg > 0:
-            gns = self._s_avg / self._g2_avg
-        self.gns_by_epoch[epoch] = gns
+        self.gns_by_epoch[epoch] = self.current_gns()
 
     def should_double(self, epoch: int, lookback: int = 10) -> bool:
         """Every ``lookback`` epochs: double bs if current GNS exceeds the
@@ -92,15 +105,19 @@ class GNSEstimator:
         return cur > sum(recent) / len(recent)
 
     def state_dict(self):
+        g2, s = (self._avg.tolist() if self._have_avg else (None, None))
         return {
-            "g2_avg": self._g2_avg,
-            "s_avg": self._s_avg,
+            "g2_avg": g2,
+            "s_avg": s,
             "gns_by_epoch": self.gns_by_epoch,
             "batch_size": self.batch_size,
         }
 
     def load_state_dict(self, state):
-        self._g2_avg = state["g2_avg"]
-        self._s_avg = state["s_avg"]
+        if state.get("g2_avg") is not None:
+            self._avg = torch.tensor(
+                [state["g2_avg"], state["s_avg"]], device=self._device
+            )
+            self._have_avg = True
         self.gns_by_epoch = {int(k): v for k, v in state["gns_by_epoch"].items()}
         self.batch_size = state["batch_size"]

@@ -1,0 +1,131 @@
+"""Adaptation-library numerics (CPU reference path)."""
+
+import math
+
+import pytest
+import torch
+
+from shockwave_amd.adapt import (
+    AccordionDetector,
+    GNSEstimator,
+    hardcoded_critical_regime,
+)
+
+
+def tiny_model():
+    torch.manual_seed(0)
+    return torch.nn.Sequential(
+        torch.nn.Conv2d(3, 4, 3), torch.nn.Flatten(), torch.nn.LazyLinear(2)
+    )
+
+
+class TestAccordion:
+    def _run_steps(self, det, model, n):
+        for _ in range(n):
+            x = torch.randn(2, 3, 8, 8)
+            model.zero_grad()
+            model(x).sum().backward()
+            det.on_step()
+
+    def test_accumulates_and_norms(self):
+        model = tiny_model()
+        model(torch.randn(1, 3, 8, 8))  # materialize lazy layer
+        det = AccordionDetector(model, interval=2)
+        self._run_steps(det, model, 3)
+        # accumulated = sum of grads over steps
+        expected = det.accum[0].clone()
+        norms = det.on_epoch(0)
+        assert det.norms_by_epoch[0][0] == pytest.approx(
+            expected.norm().item(), rel=1e-5
+        )
+        # buffers reset after epoch
+        assert all(a.abs().sum() == 0 for a in det.accum)
+
+    def test_regime_transition(self):
+        model = tiny_model()
+        model(torch.randn(1, 3, 8, 8))
+        det = AccordionDetector(model, interval=1, threshold=0.5)
+        # epoch 0: large norms; epoch 1: tiny norms -> rel change > 0.5
+        for p in det.params:
+            p.grad = torch.ones_like(p) * 10
+        det.on_step()
+        det.on_epoch(0)
+        for p in det.params:
+            p.grad = torch.ones_like(p) * 10
+        det.on_step()
+        decision = det.on_epoch(1)  # same -> rel change 0 -> leaves regime
+        assert det.in_critical_regime is False
+        for p in det.params:
+            p.grad = torch.ones_like(p) * 100
+        det.on_step()
+        decision = det.on_epoch(2)  # 10x change -> back in critical regime
+        assert det.in_critical_regime is True
+        assert decision is True
+
+    def test_hardcoded_tables(self):
+        assert hardcoded_critical_regime("ResNet-18", 32, 5)
+        assert not hardcoded_critical_regime("ResNet-18", 32, 50)
+        assert hardcoded_critical_regime("ResNet-18", 32, 155)
+        assert hardcoded_critical_regime("ResNet-50", 64, 45) is False
+        assert hardcoded_critical_regime("ResNet-50", 64, 35) is True
+        assert hardcoded_critical_regime("Transformer", 32, 999)
+
+    def test_state_roundtrip(self):
+        model = tiny_model()
+        model(torch.randn(1, 3, 8, 8))
+        det = AccordionDetector(model, interval=1)
+        for p in det.params:
+            p.grad = torch.ones_like(p)
+        det.on_step()
+        det.on_epoch(0)
+        state = det.state_dict()
+        det2 = AccordionDetector(model, interval=1)
+        det2.load_state_dict(state)
+        assert det2.norms_by_epoch == det.norms_by_epoch
+
+
+class TestGNS:
+    def test_estimates_match_manual(self):
+        model = torch.nn.Linear(8, 4)
+        est = GNSEstimator(model, batch_size=10, window=2, ema=0.0)
+        grads = []
+        for i in range(2):
+            torch.manual_seed(i)
+            model.zero_grad()
+            model(torch.randn(4, 8)).sum().backward()
+            grads.append(
+                torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+            )
+            est.on_step()
+        big = (sum(grads) / 2).norm() ** 2
+        small = grads[-1].norm() ** 2
+        b_small, b_big = 10.0, 20.0
+        g2 = (b_big * big - b_small * small) / (b_big - b_small)
+        s = (small - big) / (1 / b_small - 1 / b_big)
+        assert est.current_gns() == pytest.approx((s / g2).item(), rel=1e-4)
+
+    def test_should_double_logic(self):
+        model = torch.nn.Linear(4, 2)
+        est = GNSEstimator(model, batch_size=8, window=2)
+        for e in range(10, 19):
+            est.gns_by_epoch[e] = 1.0
+        est.gns_by_epoch[19] = 5.0  # spike above trailing average
+        assert est.should_double(19)
+        for e in range(20, 29):
+            est.gns_by_epoch[e] = 1.0
+        est.gns_by_epoch[29] = 0.5  # below trailing average
+        assert not est.should_double(29)
+        assert not est.should_double(7)   # before first decision window
+        assert not est.should_double(17)  # not a decision epoch
+
+    def test_window_copies_not_views(self):
+        """Grad buffers are reused across steps; the window must snapshot."""
+        model = torch.nn.Linear(4, 2)
+        est = GNSEstimator(model, batch_size=8, window=2)
+        for p in model.parameters():
+            p.grad = torch.ones_like(p)
+        est.on_step()
+        first = est._grads[0].clone()
+        for p in model.parameters():
+            p.grad.fill_(99.0)
+        assert torch.equal(est._grads[0], first)
