@@ -74,11 +74,14 @@ def cifar10_main(argv=None, mode=None, client=None, max_steps_override=None):
         y = y.to(device, non_blocking=True)
         return criterion(model(x), y)
 
+    import os as _os
+
+    dataset_len = int(_os.environ.get("SWQ_DATASET_LEN", 50000))
     spec = WorkloadSpec(
         family="ResNet-18",
         build_model=lambda a, d: _nhwc(resnet18_cifar().to(d), d),
         build_loader=lambda a: _loader(
-            synthetic.SyntheticImages(50000, 32, 10), a, a.batch_size
+            synthetic.SyntheticImages(dataset_len, 32, 10), a, a.batch_size
         ),
         build_optimizer=lambda a, params: FusedSGD(
             params, lr=a.lr * a.batch_size / 128, momentum=0.9,

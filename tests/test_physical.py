@@ -89,3 +89,69 @@ class TestPhysicalEndToEnd:
             )
         finally:
             sched.shutdown()
+
+
+@pytest.mark.slow
+class TestPhysicalAdaptation:
+    def test_accordion_rescale_through_full_stack(self, tmp_path, throughputs):
+        """An accordion job leaves its critical regime inside the real
+        training subprocess, reports big_bs over gRPC, the scheduler
+        rewrites the command (epoch-preserving step rescale), and the job
+        finishes at the new batch size (BASELINE config 4 core loop)."""
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        os.environ["SWQ_DATASET_LEN"] = "96"  # 3 steps/epoch at bs 32
+        try:
+            port = free_port()
+            worker_port = free_port()
+            job = Job(
+                job_id=None,
+                job_type="ResNet-18 (batch size 32)",
+                command="python3 main.py --batch_size 32",
+                working_directory="image_classification/cifar10",
+                num_steps_arg="--num_steps",
+                total_steps=60,
+                duration=600,
+                scale_factor=1,
+                mode="accordion",
+            )
+            profiles = [trace_mod.build_job_profile(job, throughputs)]
+            sched = PhysicalScheduler(
+                get_policy("max_min_fairness"),
+                port=port,
+                expected_num_workers=1,
+                throughputs=throughputs,
+                time_per_iteration=25,
+                profiles=profiles,
+                worker_type="mi355x",
+            )
+            try:
+                Worker(
+                    worker_type="mi355x",
+                    sched_addr="127.0.0.1",
+                    sched_port=port,
+                    worker_port=worker_port,
+                    num_gpus=1,
+                    ip_addr="127.0.0.1",
+                    run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                    accordion_run_dir=os.path.join(
+                        REPO, "workloads", "accordion"
+                    ),
+                    gns_run_dir=os.path.join(REPO, "workloads", "gns"),
+                    checkpoint_dir=str(tmp_path),
+                )
+                sched.add_job(job)
+                deadline = time.time() + 300
+                while not sched.is_done() and time.time() < deadline:
+                    time.sleep(2)
+                assert len(sched.get_job_completion_times()) == 1, \
+                    "accordion job did not complete"
+                # the scheduler rescaled the job to the family's max bs
+                assert job.batch_size == 256, job.command
+                assert "--batch_size 256" in job.command
+            finally:
+                sched.shutdown()
+        finally:
+            del os.environ["SWQ_DATASET_LEN"]
