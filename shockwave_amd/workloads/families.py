@@ -88,6 +88,16 @@ def cifar10_main(argv=None, mode=None, client=None, max_steps_override=None):
             weight_decay=5e-4,
         ),
         step=step,
+        make_static_batch=lambda a, d: (
+            torch.zeros(a.batch_size, 3, 32, 32, device=d).to(
+                memory_format=torch.channels_last
+            ),
+            torch.zeros(a.batch_size, dtype=torch.long, device=d),
+        ),
+        copy_batch=lambda s, b: (
+            s[0].copy_(b[0], non_blocking=True),
+            s[1].copy_(b[1], non_blocking=True),
+        ),
     )
     return run(spec, args, mode=mode, client=client,
                max_steps_override=max_steps_override)
@@ -123,6 +133,16 @@ def imagenet_main(argv=None, mode=None, client=None, max_steps_override=None):
             params, lr=a.lr, momentum=0.9, weight_decay=1e-4
         ),
         step=step,
+        make_static_batch=lambda a, d: (
+            torch.zeros(a.batch_size, 3, 224, 224, device=d).to(
+                memory_format=torch.channels_last
+            ),
+            torch.zeros(a.batch_size, dtype=torch.long, device=d),
+        ),
+        copy_batch=lambda s, b: (
+            s[0].copy_(b[0], non_blocking=True),
+            s[1].copy_(b[1], non_blocking=True),
+        ),
     )
     return run(spec, args, mode=mode, client=client,
                max_steps_override=max_steps_override)

@@ -36,6 +36,12 @@ class WorkloadSpec:
     synthetic_data: bool = True
     extra_state: Callable = None     # (state) -> dict to checkpoint
     restore_state: Callable = None   # (state, dict) -> None
+    # hipGraph support: build a device-resident batch with the loader's
+    # structure + a copier from a host batch into it.  Static-mode jobs on
+    # GPU then run one graph replay per step instead of ~400 eager
+    # launches (shockwave_amd/parallel/graphs.py).
+    make_static_batch: Callable = None   # (args, device) -> batch
+    copy_batch: Callable = None          # (static_batch, batch) -> None
 
 
 def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
@@ -138,20 +144,53 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
         return False
 
     model_train.train()
+
+    def eager_step(batch):
+        common.zero_grads(model_train)
+        loss = spec.step(model_train, batch, device, state)
+        if loss is not None and loss.requires_grad:
+            loss.backward()
+        common.finish_sync(model_train)
+        if accordion is not None:
+            accordion.on_step()
+        if gns is not None:
+            gns.on_step()
+        optimizer.step()
+
+    # hipGraph the whole compute step for static-mode GPU jobs: the
+    # adaptation hooks of accordion (multi-tensor accumulate) would be
+    # capturable, but GNS's window snapshots allocate per step, so graphs
+    # are enabled for static jobs only
+    graphed = None
+    static_batch = None
+    if (
+        device.type == "cuda"
+        and mode == "static"
+        and spec.make_static_batch is not None
+        and os.environ.get("SWQ_GRAPHS", "1") != "0"
+    ):
+        from ..parallel.graphs import try_graph_step
+
+        static_batch = spec.make_static_batch(args, device)
+
+        def graph_body():
+            common.zero_grads(model_train)
+            loss = spec.step(model_train, static_batch, device, state)
+            loss.backward()
+            common.finish_sync(model_train)
+            optimizer.step()
+
+        graphed = try_graph_step(lambda: graph_body(), [])
+
     done = False
     while not done and state["cumulative_steps"] < target_steps:
         hit_target = False
         for batch in trainloader:
-            common.zero_grads(model_train)
-            loss = spec.step(model_train, batch, device, state)
-            if loss is not None and loss.requires_grad:
-                loss.backward()
-            common.finish_sync(model_train)
-            if accordion is not None:
-                accordion.on_step()
-            if gns is not None:
-                gns.on_step()
-            optimizer.step()
+            if graphed is not None:
+                spec.copy_batch(static_batch, batch)
+                graphed.replay()
+            else:
+                eager_step(batch)
             state["cumulative_steps"] += 1
             reporter.step()
             if state["cumulative_steps"] >= target_steps:
