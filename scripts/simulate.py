@@ -121,7 +121,12 @@ def run_simulation(
 
 def main():
     p = argparse.ArgumentParser(description=__doc__)
-    p.add_argument("-t", "--trace_file", required=True)
+    p.add_argument("-t", "--trace_file", default=None)
+    p.add_argument("--generate_jobs", type=int, default=None,
+                   help="generate N jobs on the fly instead of a trace "
+                        "(reference simulate_scheduler_with_generated_jobs)")
+    p.add_argument("--lam", type=float, default=60.0,
+                   help="mean interarrival seconds for --generate_jobs")
     p.add_argument("--throughputs_file", default="traces/mi355x_throughputs.json")
     p.add_argument("-p", "--policy", default="shockwave")
     p.add_argument("-c", "--config", default=None, help="shockwave config json")
@@ -131,6 +136,27 @@ def main():
     p.add_argument("--results_dir", default="results")
     p.add_argument("--log_level", default="WARNING")
     args = p.parse_args()
+
+    if args.generate_jobs:
+        import tempfile
+
+        from shockwave_amd.core import generator
+        from shockwave_amd.core.throughputs import read_throughputs as _rt
+        from shockwave_amd.core.trace import (
+            canonical_worker_type as _cwt,
+            write_trace as _wt,
+        )
+
+        tputs = _rt(args.throughputs_file)
+        jobs, arrivals = generator.generate_trace(
+            tputs, _cwt(tputs), args.generate_jobs, lam_s=args.lam,
+            seed=args.seed,
+        )
+        fd, args.trace_file = tempfile.mkstemp(suffix=".trace")
+        os.close(fd)
+        _wt(jobs, arrivals, args.trace_file)
+        print(f"generated {len(jobs)} jobs -> {args.trace_file}")
+    assert args.trace_file, "need --trace_file or --generate_jobs"
 
     r = run_simulation(
         args.trace_file,

@@ -615,6 +615,7 @@ class PhysicalScheduler(RoundScheduler):
                 return
 
             # per-worker cumulative run time (reference :4374-4380)
+            self._cumulative_run_time.setdefault(job_id, {})
             self._cumulative_run_time[job_id].setdefault(worker_id, 0.0)
             self._cumulative_run_time[job_id][worker_id] += float(
                 np.max(all_execution_times)
@@ -641,10 +642,10 @@ class PhysicalScheduler(RoundScheduler):
             run_time_so_far = (
                 sum(self._cumulative_run_time[job_id].values()) / scale_factor
             )
-            is_over_deadline = any(
-                is_active.values()
-            ) and run_time_so_far > int(
-                self._jobs[job_id].duration * 1.5
+            lead = job_id.singletons()[0]
+            is_over_deadline = (
+                lead in self._jobs
+                and run_time_so_far > int(self._jobs[lead].duration * 1.5)
             )
 
             self._in_progress_updates[job_id].sort(key=lambda x: x[0])

@@ -611,6 +611,10 @@ class RoundScheduler:
             if name.startswith("FIFO") and priority <= 0.0:
                 continue
             if scale_factor > num_workers_left[worker_type]:
+                if name == "Isolated_plus":
+                    # isolated_plus strictly respects priority order: stop
+                    # rather than skip ahead (reference :1247-1252)
+                    break
                 continue
             num_workers_left[worker_type] -= scale_factor
             for s in singles:
@@ -938,18 +942,21 @@ class RoundScheduler:
         all_iterator_logs=None,
     ):
         to_remove = []
+        # pair (packed) job ids are keyed per combination
+        self._cumulative_run_time.setdefault(job_id, {})
         self._cumulative_run_time[job_id].setdefault(worker_id, 0.0)
         self._cumulative_run_time[job_id][worker_id] += float(
             np.max(all_execution_times)
         )
 
-        if job_id in self._jobs:
+        lead = job_id.singletons()[0]
+        if lead in self._jobs:
             run_time_so_far = (
                 sum(self._cumulative_run_time[job_id].values())
-                / self._jobs[job_id].scale_factor
+                / self._jobs[lead].scale_factor
             )
             is_over_deadline = run_time_so_far > int(
-                self._jobs[job_id].duration * 1.5
+                self._jobs[lead].duration * 1.5
             )
         else:
             is_over_deadline = True
@@ -1369,20 +1376,48 @@ class RoundScheduler:
         return (float(np.mean(utilizations)) if utilizations else 0.0, utilizations)
 
     def get_envy_list(self):
-        """Pairwise envy: how much more time another job received relative
-        to own allocation-normalized share (reference :2966-3014 in spirit:
-        completed-job envy ratios over per-round scheduled counts)."""
-        envy_ratios = []
-        sched = self._num_scheduled_rounds
-        for a in self._job_completion_times:
-            ra = sched.get(a, 0)
-            for b in self._job_completion_times:
-                if a == b:
-                    continue
-                rb = sched.get(b, 0)
-                if ra > 0:
-                    envy_ratios.append(rb / ra)
-        return envy_ratios
+        """Per-job envy ratio = scheduled_rounds / (scheduled + queued)
+        (the reciprocal of the sharing slowdown), plus all pairwise
+        absolute differences (reference get_envy_list :2966-3014)."""
+        envy_ratios = OrderedDict()
+        for job_id, sched_rounds in self._num_scheduled_rounds.items():
+            queued = self._num_queued_rounds.get(job_id, 0)
+            total = sched_rounds + queued
+            envy_ratios[job_id] = sched_rounds / total if total > 0 else 0.0
+        vals = list(envy_ratios.values())
+        vals_absdiff = [
+            abs(vi - vj)
+            for j, vj in enumerate(vals)
+            for i, vi in enumerate(vals)
+            if i > j
+        ]
+        return envy_ratios, vals_absdiff
+
+    def get_throughput_timeline(self):
+        return self._throughput_timeline
+
+    def get_job_run_time(self):
+        return self._cumulative_run_time
+
+    def get_completed_steps(self, job_ids=None):
+        if job_ids is None:
+            job_ids = sorted(self._total_steps_run.keys())
+        return {j: self._total_steps_run[j] for j in job_ids
+                if j in self._total_steps_run}
+
+    def save_job_timelines(self, timeline_dir):
+        """Dump each job's accumulated iterator logs (reference
+        save_job_timelines :3109-3128)."""
+        import os
+
+        os.makedirs(timeline_dir, exist_ok=True)
+        for job_id, per_worker in self._job_timelines.items():
+            for i, lines in enumerate(per_worker):
+                path = os.path.join(
+                    timeline_dir, f"job_id={job_id[0]}.worker={i}.log"
+                )
+                with open(path, "w") as f:
+                    f.write("\n".join(lines))
 
     def get_num_lease_extensions(self):
         if self._num_lease_extension_opportunities > 0:

@@ -91,6 +91,17 @@ class Worker:
             gns_run_dir=gns_run_dir,
         )
         self._done = threading.Event()
+        self._heartbeat_thread = threading.Thread(
+            target=self._heartbeat_loop, daemon=True
+        )
+        self._heartbeat_thread.start()
+
+    def _heartbeat_loop(self, interval_s: float = 30.0):
+        while not self._done.wait(interval_s):
+            try:
+                self._rpc_client.send_heartbeat()
+            except Exception:
+                logger.warning("heartbeat to scheduler failed", exc_info=True)
 
     # -- RPC callbacks -------------------------------------------------------
 
