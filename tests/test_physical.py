@@ -155,3 +155,72 @@ class TestPhysicalAdaptation:
                 sched.shutdown()
         finally:
             del os.environ["SWQ_DATASET_LEN"]
+
+
+@pytest.mark.slow
+class TestPhysicalDistributedJob:
+    def test_scale_factor_2_job_with_shockwave_policy(self, tmp_path,
+                                                      throughputs):
+        """A 2-GPU data-parallel job: the scheduler assigns two worker ids,
+        injects --master_addr/--master_port/--world_size/--rank at
+        dispatch, both ranks rendezvous (gloo on CPU), renew leases via
+        the first-requester path, and the job completes under the
+        shockwave planner."""
+        import json
+
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        port = free_port()
+        worker_port = free_port()
+        job = Job(
+            job_id=None,
+            job_type="ResNet-18 (batch size 16)",
+            command="python3 main.py --batch_size 16",
+            working_directory="image_classification/cifar10",
+            num_steps_arg="--num_steps",
+            total_steps=12,
+            duration=600,
+            scale_factor=2,
+            mode="static",
+        )
+        profiles = [trace_mod.build_job_profile(job, throughputs)]
+        shockwave_config = {
+            "future_rounds": 5, "k": 1e-3, "lambda": 12.0, "rhomax": 1.0,
+            "time_per_iteration": 30, "num_gpus": 2,
+        }
+        sched = PhysicalScheduler(
+            get_policy("shockwave"),
+            port=port,
+            expected_num_workers=2,
+            throughputs=throughputs,
+            time_per_iteration=30,
+            profiles=profiles,
+            shockwave_config=shockwave_config,
+            worker_type="mi355x",
+        )
+        try:
+            Worker(
+                worker_type="mi355x",
+                sched_addr="127.0.0.1",
+                sched_port=port,
+                worker_port=worker_port,
+                num_gpus=2,
+                ip_addr="127.0.0.1",
+                run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                accordion_run_dir=os.path.join(REPO, "workloads", "accordion"),
+                gns_run_dir=os.path.join(REPO, "workloads", "gns"),
+                checkpoint_dir=str(tmp_path),
+            )
+            sched.add_job(job)
+            deadline = time.time() + 300
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(2)
+            completions = sched.get_job_completion_times()
+            assert len(completions) == 1, "2-GPU job did not complete"
+            # both ranks contributed steps
+            steps = sched.get_completed_steps()
+            assert list(steps.values())[0] >= 12
+        finally:
+            sched.shutdown()
