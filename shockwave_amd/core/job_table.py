@@ -10,7 +10,7 @@ directory at dispatch time when ``needs_data_dir`` is set.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass(frozen=True)

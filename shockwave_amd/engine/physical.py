@@ -20,7 +20,6 @@ Extends the RoundScheduler state machine with the live control plane
 
 from __future__ import annotations
 
-import collections
 import copy
 import logging
 import math
@@ -32,7 +31,6 @@ from concurrent.futures import ThreadPoolExecutor
 
 import numpy as np
 
-from ..core.job import JobIdPair
 from ..rpc.services import SchedulerRpcClient, serve_scheduler
 from ..runtime.set_queue import SetQueue
 from .scheduler import (

@@ -7,8 +7,6 @@ step instead of per-parameter elementwise ops.
 
 from __future__ import annotations
 
-from typing import Iterable
-
 import torch
 
 from . import fused_adam, fused_sgd

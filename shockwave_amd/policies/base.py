@@ -14,7 +14,7 @@ constraints (policy.py:59-65):
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import numpy as np
 from scipy.optimize import linprog

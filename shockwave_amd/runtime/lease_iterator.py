@@ -31,8 +31,6 @@ import logging
 import os
 import time
 from collections.abc import Iterable
-from typing import Optional
-
 import torch
 from filelock import FileLock
 

@@ -13,8 +13,6 @@ from __future__ import annotations
 import argparse
 import os
 import time
-from typing import Callable, Optional
-
 import torch
 import torch.distributed as dist
 
