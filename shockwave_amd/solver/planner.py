@@ -250,7 +250,12 @@ class ShockwavePlanner:
             ratio = projected / bound if bound > 0 else 1.0
             if ratio > self.rhomax:
                 power = priority_M if remaining < self.round_duration else self.lam
-                priorities.append(ratio ** power)
+                # ratio**power in log space: Python floats raise
+                # OverflowError past 1e308 (large rho with power=100)
+                import math
+
+                log_p = power * math.log(max(ratio, 1e-9))
+                priorities.append(math.exp(min(log_p, 700.0)))
             else:
                 priorities.append(1.0)
         # normalize giant priorities to avoid numeric blowup in the MILP

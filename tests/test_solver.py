@@ -156,3 +156,30 @@ class TestPlanner:
         assert p.resolve
         sched = p.round_schedule()
         assert 0 not in sched
+
+
+class TestRelaxedPriorities:
+    def test_huge_rho_no_overflow(self, throughputs):
+        """ratio**lambda with rho >> 1 must not overflow (hit on GPU when a
+        tiny job's ftf bound was long passed)."""
+        from tests.test_core import make_job
+        from shockwave_amd.core import trace
+
+        md = OrderedDict()
+        j = make_job(mode="static", steps=1563 * 5)
+        prof = trace.build_job_profile(j, throughputs)
+        m = JobMetadata(0, prof)
+        m.register_submit(0.0)
+        md[0] = m
+        p = ShockwavePlanner(
+            ngpus=1, gram=288, init_metadata=md, future_nrounds=3,
+            round_duration=30,
+        )
+        # fabricate a stale tiny bound so projected/bound explodes
+        p.finish_time_uniform_share()
+        p.share_series[0] = [(0, 1e-6)]
+        p.round_ptr = 1000
+        pri = p._relaxed_priorities([0])
+        assert all(np.isfinite(pri))
+        sched = p._greedy_schedule([0])
+        assert sched.shape == (1, 3)
