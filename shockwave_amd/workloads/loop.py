@@ -120,9 +120,9 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
         lease_it.save_checkpoint(out)
 
     def maybe_request_rescale(epoch):
-        """Epoch-boundary batch-size decisions (shared across families)."""
-        if lease_it is None:
-            return False
+        """Epoch-boundary adaptation: always run the detectors (per-layer
+        norms / GNS read); report a batch-size change only when scheduled
+        under a lease."""
         bs = getattr(args, "batch_size", 0)
         max_bs = ds_tables.max_batch_size(spec.family, bs)
         if accordion is not None:
@@ -130,15 +130,20 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
             in_cr = hardcoded_critical_regime(
                 spec.family, state["original_bs"], epoch + 1
             )
-            if not in_cr and bs == state["original_bs"] and bs != max_bs:
-                lease_it.update_resource_requirement(True, False)
-                return True
-            if in_cr and bs != state["original_bs"]:
-                lease_it.update_resource_requirement(False, True)
-                return True
+            if lease_it is not None:
+                if not in_cr and bs == state["original_bs"] and bs != max_bs:
+                    lease_it.update_resource_requirement(True, False)
+                    return True
+                if in_cr and bs != state["original_bs"]:
+                    lease_it.update_resource_requirement(False, True)
+                    return True
         if gns is not None:
             gns.on_epoch(epoch)
-            if gns.should_double(epoch) and bs < max_bs:
+            if (
+                lease_it is not None
+                and gns.should_double(epoch)
+                and bs < max_bs
+            ):
                 lease_it.update_resource_requirement(True, False)
                 return True
         return False
