@@ -20,8 +20,9 @@ class SyntheticImages(Dataset):
         self.image_size = image_size
         self.num_classes = num_classes
         g = torch.Generator().manual_seed(seed)
-        # small resident pool re-indexed modulo, so memory stays bounded
-        pool = min(num_samples, 2048)
+        # small resident pool re-indexed modulo, so memory (and per-job
+        # startup time) stays bounded: 2048 images at 32x32, 256 at 224x224
+        pool = min(num_samples, 2048 if image_size <= 64 else 256)
         self.images = torch.randn(pool, 3, image_size, image_size, generator=g)
         self.labels = torch.randint(
             0, num_classes, (pool,), generator=g

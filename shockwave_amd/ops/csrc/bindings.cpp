@@ -56,8 +56,17 @@ void check_lists(const std::vector<std::vector<torch::Tensor>>& lists) {
                         "tensor ", t, " not dense");
             TORCH_CHECK(ten.numel() == lists[0][t].numel(),
                         "numel mismatch across lists at tensor ", t);
-            TORCH_CHECK(ten.strides() == lists[0][t].strides(),
-                        "stride/layout mismatch across lists at tensor ", t);
+            // layouts must agree so the flat kernels pair elements
+            // correctly; strides of size-1 dims are don't-care (a 1x1
+            // conv's NCHW-contiguous grad is layout-equal to its
+            // channels_last param even though the stride tuples differ)
+            const auto& ref = lists[0][t];
+            for (int64_t d = 0; d < ten.dim(); ++d) {
+                if (ten.size(d) <= 1) continue;
+                TORCH_CHECK(ten.stride(d) == ref.stride(d),
+                            "stride/layout mismatch across lists at tensor ",
+                            t, " dim ", d);
+            }
         }
     }
 }

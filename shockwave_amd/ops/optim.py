@@ -12,12 +12,26 @@ import torch
 from . import fused_adam, fused_sgd
 
 
+def _preinit_grads(param_groups):
+    """Create zero gradients with each param's own memory layout before
+    the first backward: autograd then accumulates into them, so grad
+    layout always matches the param (a backward kernel is otherwise free
+    to emit a different-layout grad, e.g. NCHW grads for channels_last
+    1x1-conv weights) and grad addresses stay stable for the metadata
+    cache and hipGraphs."""
+    for group in param_groups:
+        for p in group["params"]:
+            if p.requires_grad and p.grad is None:
+                p.grad = torch.zeros_like(p)
+
+
 class FusedSGD(torch.optim.Optimizer):
     def __init__(self, params, lr, momentum=0.0, dampening=0.0,
                  weight_decay=0.0, nesterov=False):
         defaults = dict(lr=lr, momentum=momentum, dampening=dampening,
                         weight_decay=weight_decay, nesterov=nesterov)
         super().__init__(params, defaults)
+        _preinit_grads(self.param_groups)
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -58,6 +72,7 @@ class FusedAdam(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay, adamw=adamw)
         super().__init__(params, defaults)
+        _preinit_grads(self.param_groups)
 
     @torch.no_grad()
     def step(self, closure=None):
