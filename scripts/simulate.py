@@ -32,6 +32,7 @@ def run_simulation(
     shockwave_config_file=None,
     results_dir=None,
     log_level="WARNING",
+    preemption_overhead_s=20.0,
 ):
     import logging
 
@@ -70,6 +71,7 @@ def run_simulation(
         profiles=profiles,
         shockwave_config=shockwave_config,
         worker_type=worker_type,
+        preemption_overhead_s=preemption_overhead_s,
     )
 
     start = time.time()
@@ -135,6 +137,8 @@ def main():
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--results_dir", default="results")
     p.add_argument("--log_level", default="WARNING")
+    p.add_argument("--preemption_overhead", type=float, default=20.0,
+                   help="simulated checkpoint/restart cost per migration (s)")
     args = p.parse_args()
 
     if args.generate_jobs:
@@ -168,6 +172,7 @@ def main():
         shockwave_config_file=args.config,
         results_dir=args.results_dir,
         log_level=args.log_level,
+        preemption_overhead_s=args.preemption_overhead,
     )
     print(
         json.dumps(

@@ -69,6 +69,7 @@ class RoundScheduler:
         profiles: Optional[List[Dict]] = None,
         shockwave_config: Optional[Dict] = None,
         worker_type: str = "mi355x",
+        preemption_overhead_s: float = PREEMPTION_OVERHEAD_S,
     ):
         self._policy = policy
         self._simulate = simulate
@@ -80,6 +81,7 @@ class RoundScheduler:
         self._max_rounds = max_rounds
         self._profiles = profiles
         self._worker_type = worker_type
+        self._preemption_overhead_s = preemption_overhead_s
         self._job_packing = "Packing" in getattr(policy, "name", "")
 
         self._start_timestamp = 0.0 if simulate else time.time()
@@ -735,28 +737,39 @@ class RoundScheduler:
     # Step accounting (reference :1425-1516)
     # ------------------------------------------------------------------
 
-    def _get_num_steps(self, job_id, worker_type, single_job_id=None):
+    def _get_num_steps(self, job_id, worker_type, single_job_id=None,
+                       startup_s=0.0):
+        effective_time = max(0.0, self._time_per_iteration - startup_s)
         if self._simulate and job_id.is_pair():
             assert single_job_id is not None
             index = 0 if job_id.singletons()[0] == single_job_id else 1
             num_steps = int(
-                self._throughputs[job_id][worker_type][index]
-                * self._time_per_iteration
+                self._throughputs[job_id][worker_type][index] * effective_time
             )
         else:
             tput = self._throughputs[job_id][worker_type]
             if job_id.is_pair():
                 index = 0 if job_id.singletons()[0] == single_job_id else 1
                 tput = tput[index]
-            num_steps = int(tput * self._time_per_iteration)
+            num_steps = int(tput * effective_time)
         target = single_job_id if single_job_id is not None else job_id
         return min(num_steps, self._get_remaining_steps(target))
 
-    def _get_job_steps_and_finish_times(self, job_id, worker_type):
+    def _get_job_steps_and_finish_times(self, job_id, worker_type,
+                                        startup_s=0.0):
+        """Steps achievable this round and the finish time.  A job newly
+        placed on its workers pays ``startup_s`` of checkpoint-restore /
+        process-startup cost before training resumes (the reference clips
+        a 20 s penalty off full-round micro-tasks at completion instead,
+        scheduler.py:1936-1968, which under-counts for jobs shorter than a
+        round; charging it at dispatch matches the physical mechanism,
+        where every dispatch launches a fresh process)."""
         max_finish_time = self.get_current_timestamp()
         all_num_steps = []
         for single in job_id.singletons():
-            num_steps = self._get_num_steps(job_id, worker_type, single)
+            num_steps = self._get_num_steps(
+                job_id, worker_type, single, startup_s=startup_s
+            )
             all_num_steps.append(num_steps)
             tput = self._throughputs[job_id][worker_type]
             if job_id.is_pair():
@@ -766,7 +779,9 @@ class RoundScheduler:
                 raise RuntimeError(
                     f"throughput for job {single} on {worker_type} <= 0"
                 )
-            finish_time = self.get_current_timestamp() + num_steps / tput
+            finish_time = (
+                self.get_current_timestamp() + startup_s + num_steps / tput
+            )
             max_finish_time = max(max_finish_time, finish_time)
             self._running_jobs.add(single)
         return all_num_steps, max_finish_time
@@ -1153,28 +1168,12 @@ class RoundScheduler:
                 if finish_time > self._current_timestamp:
                     break
                 all_execution_times = []
-                nfs_slowdown_factor = 1
                 for single in job_id.singletons():
                     execution_time = finish_time - current_round_start_time
-                    if current_round != 1 and current_round >= 2:
-                        prev_sched = self._per_round_schedule[current_round - 2]
-                        if single[0] not in prev_sched:
-                            # migration: inject checkpoint/restore overhead
-                            if (
-                                execution_time != 0
-                                and self._time_per_iteration - 5 < execution_time
-                            ):
-                                nfs_slowdown_factor = (
-                                    execution_time - PREEMPTION_OVERHEAD_S
-                                ) / execution_time
-                                execution_time -= PREEMPTION_OVERHEAD_S
                     all_execution_times.append(execution_time)
                     self._per_job_latest_timestamps[single] = finish_time
                 self._in_progress_updates[job_id] = []
                 scale_factor = self._jobs[job_id.singletons()[0]].scale_factor
-                all_num_steps = [
-                    int(x * nfs_slowdown_factor) for x in all_num_steps
-                ]
                 total_steps = [0] * len(job_id.singletons())
                 for i, worker_id in enumerate(worker_ids):
                     if i == len(worker_ids) - 1:
@@ -1257,12 +1256,23 @@ class RoundScheduler:
                         self._num_lease_extensions += 1
             self._current_worker_assignments = scheduled_jobs
 
+            prev_round_jobs = (
+                set(self._per_round_schedule[-2].keys())
+                if len(self._per_round_schedule) >= 2
+                else set()
+            )
             for job_id, worker_ids in scheduled_jobs.items():
                 worker_type = self._worker_id_to_worker_type_mapping[
                     worker_ids[0]
                 ]
+                newly_placed = job_id[0] not in prev_round_jobs
                 all_num_steps, max_finish_time = (
-                    self._get_job_steps_and_finish_times(job_id, worker_type)
+                    self._get_job_steps_and_finish_times(
+                        job_id, worker_type,
+                        startup_s=(
+                            self._preemption_overhead_s if newly_placed else 0.0
+                        ),
+                    )
                 )
                 heapq.heappush(
                     running_jobs,
