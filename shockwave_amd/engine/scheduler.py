@@ -70,6 +70,9 @@ class RoundScheduler:
         shockwave_config: Optional[Dict] = None,
         worker_type: str = "mi355x",
         preemption_overhead_s: float = PREEMPTION_OVERHEAD_S,
+        estimate_throughputs: bool = False,
+        profiling_percentage: float = 1.0,
+        num_reference_models: int = 16,
     ):
         self._policy = policy
         self._simulate = simulate
@@ -83,6 +86,29 @@ class RoundScheduler:
         self._worker_type = worker_type
         self._preemption_overhead_s = preemption_overhead_s
         self._job_packing = "Packing" in getattr(policy, "name", "")
+
+        # colocation-throughput estimation for unprofiled job types
+        # (reference scheduler.py:718-722 + throughput_estimator.py): match
+        # each new job to a reference type by ALS matrix completion and
+        # price its pairings with the reference type's pairwise profile
+        self._estimate_throughputs = estimate_throughputs and self._job_packing
+        self._reference_job_map: Dict = {}
+        self._throughput_estimator = None
+        if self._estimate_throughputs:
+            from ..core.throughput_estimator import ThroughputEstimator
+
+            job_types = [
+                k for k in throughputs[worker_type] if k[1] == 1
+            ]
+            self._throughput_estimator = ThroughputEstimator(
+                throughputs,
+                [worker_type],
+                job_types,
+                num_reference_job_types=min(num_reference_models,
+                                            len(job_types)),
+                profiling_percentage=profiling_percentage,
+                seed=seed,
+            )
 
         self._start_timestamp = 0.0 if simulate else time.time()
         self._current_timestamp = self._start_timestamp
@@ -258,6 +284,10 @@ class RoundScheduler:
                 }
             key = (job.job_type, job.scale_factor)
             other_key = (other_job.job_type, other_job.scale_factor)
+            if self._estimate_throughputs:
+                # price the pairing with the matched reference types
+                key = self._reference_job_map.get(job_id, key)
+                other_key = self._reference_job_map.get(other_job_id, other_key)
             oracle = self._oracle_throughputs[worker_type]
             if key in oracle and other_key in oracle.get(key, {}):
                 pair = oracle[key][other_key]
@@ -286,6 +316,15 @@ class RoundScheduler:
         self._num_failures_per_job[job_id] = 0
         self._total_steps_run[job_id] = 0
         self._cumulative_run_time[job_id] = {}
+        if self._estimate_throughputs and job.scale_factor == 1:
+            # pairwise colocation profiles exist for single-GPU jobs only
+            # (the reference packs equal-scale-factor jobs and its oracle
+            # carries sf=1 pairings)
+            self._reference_job_map[job_id] = (
+                self._throughput_estimator.match_job_to_reference_job(
+                    (job.job_type, 1)
+                )
+            )
         for worker_type in self._worker_types:
             self._steps_run_so_far[job_id][worker_type] = 0
             self._set_initial_throughput(job_id, worker_type)

@@ -176,3 +176,27 @@ class TestEnvyMetric:
         assert len(ratios) == 8
         assert all(0 <= v <= 1 for v in ratios.values())
         assert len(absdiff) == 8 * 7 // 2
+
+
+class TestEstimatedThroughputPacking:
+    def test_gandiva_with_estimated_throughputs(self, throughputs):
+        """Packing with colocation prices from the matrix-completion
+        estimator instead of exact pairwise profiles."""
+        import sys, os
+        sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..",
+                                        "scripts"))
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+        sched = RoundScheduler(
+            get_policy("gandiva"), simulate=True, throughputs=throughputs,
+            time_per_iteration=120, profiles=profiles, worker_type="mi355x",
+            estimate_throughputs=True, profiling_percentage=0.5,
+        )
+        makespan = sched.simulate({"mi355x": 2}, arrivals, jobs)
+        assert makespan > 0
+        assert len(sched.get_job_completion_times()) == 8
+        assert len(sched._reference_job_map) > 0
