@@ -110,3 +110,28 @@ class TestLeaseIntegration:
             SyntheticImages.__len__ = orig
         assert client.rr_calls, "accordion should have requested a rescale"
         assert client.rr_calls[0] == (True, False)
+
+
+class TestGNSLoopIntegration:
+    def test_gns_double_requested_when_estimator_fires(self, tmp_path,
+                                                       monkeypatch):
+        """When GNS says double, the loop reports big_bs and checkpoints."""
+        from tests.test_rpc_runtime import FakeLeaseClient
+        from shockwave_amd.adapt.gns import GNSEstimator
+
+        monkeypatch.setattr(GNSEstimator, "should_double",
+                            lambda self, epoch, lookback=10: True)
+        import os
+        os.environ["SWQ_DATASET_LEN"] = "64"  # 4 steps/epoch at bs16
+        try:
+            client = FakeLeaseClient([(int(1e9), 1e9)])
+            steps = families.cifar10_main(
+                ["--batch_size", "16", "--num_steps", "50",
+                 "--checkpoint_dir", str(tmp_path),
+                 "--enable_gavel_iterator", "--mode", "gns"],
+                client=client,
+            )
+            assert client.rr_calls == [(True, False)]
+            assert steps == 4  # stopped at the first epoch boundary
+        finally:
+            del os.environ["SWQ_DATASET_LEN"]
