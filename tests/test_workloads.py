@@ -132,6 +132,9 @@ class TestGNSLoopIntegration:
                 client=client,
             )
             assert client.rr_calls == [(True, False)]
-            assert steps == 4  # stopped at the first epoch boundary
+            # synthetic-data mode ends the epoch by raising on the 4th
+            # __next__ (after counting it), so the loop body ran 3 times —
+            # reference GavelIterator semantics (gavel_iterator.py:160-173)
+            assert steps == 3
         finally:
             del os.environ["SWQ_DATASET_LEN"]
