@@ -146,3 +146,31 @@ class TestUpdateLease:
             jid, 0, steps=40, duration=40.0, max_steps=100, max_duration=60
         )
         assert sched._steps_run_in_current_lease[jid] == 80  # aggregated
+
+
+class TestExtendedLeaseWatchdog:
+    def test_responsive_job_completed(self, sched, throughputs):
+        """An extended-lease job that requested renewals this round is
+        marked complete by the health check (reference :4283-4339)."""
+        jid, _ = add_job(sched, throughputs)
+        sched._current_worker_assignments = OrderedDict({jid: (0,)})
+        sched._jobs_with_extended_lease.add(jid)
+        sched._lease_update_requests[jid] = [(10, 5.0, 100, 60.0)]
+        sched._completion_events[jid] = object()
+        sched._done_callback_extended_lease(jid)
+        assert jid in sched._completed_jobs_in_current_round
+        assert jid not in sched._completion_events
+
+    def test_unresponsive_job_killed(self, sched, throughputs):
+        """No lease renewals during the round -> declared unresponsive and
+        killed; with no live workers the kill degrades to a warning."""
+        jid, _ = add_job(sched, throughputs)
+        sched._current_worker_assignments = OrderedDict({jid: (0,)})
+        sched._jobs_with_extended_lease.add(jid)
+        sched._lease_update_requests[jid] = []
+
+        killed = []
+        sched._kill_job = lambda j: killed.append(j)
+        sched._done_callback_extended_lease(jid)
+        assert killed == [jid]
+        assert jid not in sched._completed_jobs_in_current_round
