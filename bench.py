@@ -131,6 +131,8 @@ def main():
         return loss
 
     graphed = None
+    graph_mode = "eager"
+    partial = None
     if use_cuda and not args.no_graphs:
         from shockwave_amd.parallel.graphs import try_graph_step
 
@@ -139,6 +141,26 @@ def main():
             static_x = static_x.to(memory_format=torch.channels_last)
         static_y = torch.zeros(bs, dtype=torch.long, device=device)
         graphed = try_graph_step(compute_step, [static_x, static_y])
+        if graphed is not None:
+            graph_mode = "full"
+        elif distributed:
+            # RCCL collectives may not be capturable in a hipGraph on this
+            # stack: fall back to capturing fwd+bwd only and running the
+            # bucketed all-reduce + fused optimizer step eagerly (~4
+            # launches/step instead of ~400)
+            from shockwave_amd.parallel import BucketedDataParallel
+
+            assert isinstance(model, BucketedDataParallel)
+            model.sync_mode = "manual"
+
+            def fwd_bwd(x, y):
+                common.zero_grads(model)
+                loss = criterion(model(x), y)
+                loss.backward()
+
+            partial = try_graph_step(fwd_bwd, [static_x, static_y])
+            if partial is not None:
+                graph_mode = "partial"
 
     def train_step():
         x, y = next_batch()
@@ -148,6 +170,10 @@ def main():
         y = y.to(device, non_blocking=True)
         if graphed is not None:
             graphed(x, y)
+        elif partial is not None:
+            partial(x, y)
+            model.finish_gradient_sync()
+            opt.step()
         else:
             compute_step(x, y)
 

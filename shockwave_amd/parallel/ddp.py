@@ -95,10 +95,21 @@ class BucketedDataParallel(torch.nn.Module):
             self._param_bucket[p] = bucket
         self.buckets.append(bucket)
 
+    # "hook": each bucket all-reduces asynchronously as soon as its last
+    # gradient lands (overlaps with the rest of backward).  "manual": the
+    # hooks only count; finish_gradient_sync() issues every bucket's
+    # all-reduce afterwards — used when backward is hipGraph-captured but
+    # the RCCL collective must stay outside the graph.
+    sync_mode = "hook"
+
     def _grad_ready_hook(self, param):
         bucket = self._param_bucket[param]
         bucket.pending -= 1
-        if bucket.pending == 0 and self.world_size > 1:
+        if (
+            bucket.pending == 0
+            and self.world_size > 1
+            and self.sync_mode == "hook"
+        ):
             bucket.work = dist.all_reduce(
                 bucket.buffer, op=dist.ReduceOp.SUM, group=self.group,
                 async_op=True,
@@ -114,21 +125,35 @@ class BucketedDataParallel(torch.nn.Module):
 
     def finish_gradient_sync(self):
         """Call after backward, before optimizer.step()."""
-        if not self._require_finish:
-            return
+        # manual mode is driven externally (e.g. after a hipGraph replay,
+        # where the python-side forward/hooks did not re-run)
+        if self.sync_mode != "manual":
+            if not self._require_finish:
+                return
         self._require_finish = False
         if self.world_size > 1:
-            for b in self.buckets:
-                if b.pending == 0 and b.work is not None:
-                    b.work.wait()
-                elif b.pending > 0 and any(
-                    p.grad is not None for p in b.params
-                ):
-                    # partial bucket (some params unused this step): reduce now
+            if self.sync_mode == "manual":
+                works = [
                     dist.all_reduce(
-                        b.buffer, op=dist.ReduceOp.SUM, group=self.group
+                        b.buffer, op=dist.ReduceOp.SUM, group=self.group,
+                        async_op=True,
                     )
-                b.work = None
+                    for b in self.buckets
+                ]
+                for w in works:
+                    w.wait()
+            else:
+                for b in self.buckets:
+                    if b.pending == 0 and b.work is not None:
+                        b.work.wait()
+                    elif b.pending > 0 and any(
+                        p.grad is not None for p in b.params
+                    ):
+                        # partial bucket (unused params): reduce now
+                        dist.all_reduce(
+                            b.buffer, op=dist.ReduceOp.SUM, group=self.group
+                        )
+                    b.work = None
             if self.average:
                 scale = 1.0 / self.world_size
                 torch._foreach_mul_(

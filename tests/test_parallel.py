@@ -121,3 +121,26 @@ class TestBucketViews:
             )
         ddp.zero_grad()
         assert all(b.abs().sum() == 0 for b in ddp.grad_buffers)
+
+
+class TestManualSyncMode:
+    def test_manual_sync_world1(self):
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            store = dist.TCPStore("127.0.0.1", 0, 1, True)
+            dist.init_process_group("gloo", store=store, rank=0, world_size=1)
+        from shockwave_amd.parallel import BucketedDataParallel
+
+        torch.manual_seed(3)
+        model = torch.nn.Linear(8, 4)
+        ddp = BucketedDataParallel(model, bucket_bytes=64)
+        ddp.sync_mode = "manual"
+        x = torch.randn(4, 8)
+        ddp(x).sum().backward()
+        g_before = [b.clone() for b in ddp.grad_buffers]
+        ddp.finish_gradient_sync()  # world 1: average is a no-op
+        for b, g in zip(ddp.grad_buffers, g_before):
+            torch.testing.assert_close(b, g)
+        # calling again (replay-driven cadence) must not error
+        ddp.finish_gradient_sync()
