@@ -3,6 +3,12 @@
 Drop-in for the reference workloads' ``optim.SGD(momentum=0.9, wd=5e-4)``
 and Adam optimizers: one multi-tensor kernel per
 step instead of per-parameter elementwise ops.
+
+The per-group tensor lists are assembled once and cached: gradients are
+pre-created with the param's layout and never set to None (see
+``_preinit_grads`` and workloads.common.zero_grads), so the param/grad/
+state tensor identities are stable across steps — the cache removes
+~100 us of per-step Python when the GPU step itself is ~2 ms.
 """
 
 from __future__ import annotations
@@ -32,37 +38,55 @@ class FusedSGD(torch.optim.Optimizer):
                         weight_decay=weight_decay, nesterov=nesterov)
         super().__init__(params, defaults)
         _preinit_grads(self.param_groups)
+        self._cached_lists = None
+        self._buffers_initialized = False
 
-    @torch.no_grad()
-    def step(self, closure=None):
-        loss = closure() if closure is not None else None
+    def _build_lists(self):
+        cached = []
         for group in self.param_groups:
             params, grads, bufs = [], [], []
             momentum = group["momentum"]
-            buf_initialized = True
             for p in group["params"]:
                 if p.grad is None:
                     continue
                 state = self.state[p]
                 if momentum != 0 and "momentum_buffer" not in state:
                     state["momentum_buffer"] = torch.zeros_like(p)
-                    buf_initialized = False
                 params.append(p)
                 grads.append(p.grad)
                 bufs.append(
                     state["momentum_buffer"] if momentum != 0 else p.grad
                 )
+            cached.append((group, params, grads, bufs))
+        return cached
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        if self._cached_lists is None:
+            self._cached_lists = self._build_lists()
+        buf_initialized = self._buffers_initialized
+        self._buffers_initialized = True
+        for group, params, grads, bufs in self._cached_lists:
             if not params:
                 continue
             fused_sgd(
                 params, grads, bufs,
-                lr=group["lr"], momentum=momentum,
+                lr=group["lr"], momentum=group["momentum"],
                 dampening=group["dampening"],
                 weight_decay=group["weight_decay"],
                 nesterov=group["nesterov"],
                 buf_initialized=buf_initialized,
             )
         return loss
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        self._cached_lists = None  # state tensors were replaced
+        # restored momentum buffers must accumulate, not be overwritten
+        self._buffers_initialized = any(
+            "momentum_buffer" in s for s in self.state.values()
+        )
 
 
 class FusedAdam(torch.optim.Optimizer):
@@ -73,13 +97,13 @@ class FusedAdam(torch.optim.Optimizer):
                         weight_decay=weight_decay, adamw=adamw)
         super().__init__(params, defaults)
         _preinit_grads(self.param_groups)
+        self._cached_lists = None
+        self._step_count = 0
 
-    @torch.no_grad()
-    def step(self, closure=None):
-        loss = closure() if closure is not None else None
+    def _build_lists(self):
+        cached = []
         for group in self.param_groups:
             params, grads, avgs, sqs = [], [], [], []
-            step = None
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -87,23 +111,40 @@ class FusedAdam(torch.optim.Optimizer):
                 if "exp_avg" not in state:
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
-                    state["step"] = 0
-                state["step"] += 1
-                step = state["step"]
                 params.append(p)
                 grads.append(p.grad)
                 avgs.append(state["exp_avg"])
                 sqs.append(state["exp_avg_sq"])
+            cached.append((group, params, grads, avgs, sqs))
+        return cached
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        if self._cached_lists is None:
+            self._cached_lists = self._build_lists()
+        self._step_count += 1
+        for group, params, grads, avgs, sqs in self._cached_lists:
             if not params:
                 continue
             beta1, beta2 = group["betas"]
             fused_adam(
                 params, grads, avgs, sqs,
                 lr=group["lr"], beta1=beta1, beta2=beta2, eps=group["eps"],
-                weight_decay=group["weight_decay"], step=step,
+                weight_decay=group["weight_decay"], step=self._step_count,
                 adamw=group["adamw"],
             )
         return loss
+
+    def state_dict(self):
+        d = super().state_dict()
+        d["swq_step_count"] = self._step_count
+        return d
+
+    def load_state_dict(self, state_dict):
+        self._step_count = state_dict.pop("swq_step_count", 0)
+        super().load_state_dict(state_dict)
+        self._cached_lists = None
 
 
 class FusedAdamW(FusedAdam):
