@@ -8,6 +8,7 @@ from .max_min_fairness import (
     MaxMinFairnessPolicyWithPerf,
     MaxMinFairnessWaterFillingPolicy,
     MaxMinFairnessStrategyProofPolicy,
+    MaxMinFairnessStrategyProofPolicyWithPerf,
 )
 from .finish_time_fairness import (
     FinishTimeFairnessPolicy,
@@ -48,6 +49,7 @@ def get_policy(policy_name: str, seed=None, solver=None,
         "max_min_fairness_perf": MaxMinFairnessPolicyWithPerf,
         "max_min_fairness_water_filling": MaxMinFairnessWaterFillingPolicy,
         "max_min_fairness_strategy_proof": MaxMinFairnessStrategyProofPolicy,
+        "max_min_fairness_strategy_proof_perf": MaxMinFairnessStrategyProofPolicyWithPerf,
         "max_sum_throughput_perf": ThroughputSumWithPerf,
         "max_sum_throughput_normalized_by_cost_perf": ThroughputNormalizedByCostSumWithPerf,
         "max_sum_throughput_normalized_by_cost_perf_SLOs": ThroughputNormalizedByCostSumWithPerfSLOs,

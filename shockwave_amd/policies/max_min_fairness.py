@@ -189,9 +189,90 @@ class MaxMinFairnessWaterFillingPolicy(Policy):
 
 
 class MaxMinFairnessStrategyProofPolicy(MaxMinFairnessPolicy):
-    """Strategy-proof variant: LAS allocation computed on reported
-    throughputs but normalized shares use unit demands, which removes the
-    incentive to misreport (reference max_min_fairness_strategy_proof.py).
-    With a single homogeneous worker type this coincides with LAS."""
+    """Strategy-proof LAS: identical to MaxMinFairness with unit
+    throughputs — exactly what the reference's non-perf variant computes
+    (max_min_fairness_strategy_proof.py:13-45 replaces every throughput
+    with 1.0 before delegating)."""
 
     name = "MaxMinFairness_StrategyProof"
+
+
+class MaxMinFairnessStrategyProofPolicyWithPerf(Policy):
+    """Nash-bargaining allocation with VCG-style leave-one-out discounts
+    (reference max_min_fairness_strategy_proof.py:47-155): maximize the
+    geometric mean of weighted effective rates, then scale each job's
+    allocation by the product of externalities it imposes on the others
+    (rate-with-me / rate-without-me), which removes the incentive to
+    misreport throughputs.
+
+    The reference solves the geo-mean program with cvxpy; on a homogeneous
+    cluster the KKT conditions give the closed form
+    ``x_i = min(1, nu / sf_i)`` with nu found by bisection on the capacity
+    constraint (rates w_i x_i; the log objective makes x independent of
+    w beyond the weighting of the discounts)."""
+
+    name = "MaxMinFairness_Perf_StrategyProof"
+
+    def __init__(self):
+        self._proportional = ProportionalPolicy()
+
+    def _nash_x(self, sfa_col, num_workers):
+        lo, hi = 0.0, max(num_workers, sfa_col.max()) + 1.0
+        for _ in range(60):
+            nu = 0.5 * (lo + hi)
+            used = np.minimum(1.0, nu / sfa_col) * sfa_col
+            if used.sum() > num_workers:
+                hi = nu
+            else:
+                lo = nu
+        return np.minimum(1.0, lo / sfa_col)
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+        recurse_deeper=True,
+    ):
+        throughputs, index = self.flatten(unflattened_throughputs, cluster_spec)
+        if throughputs is None:
+            return None
+        m, n = throughputs.shape
+        job_ids, worker_types = index
+        assert n == 1, "strategy-proof perf assumes a homogeneous cluster"
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+
+        if recurse_deeper:
+            rates_minus_job = []
+            for jid in job_ids:
+                minus = dict(unflattened_throughputs)
+                del minus[jid]
+                rates_minus_job.append(
+                    self.get_allocation(
+                        minus, scale_factors, unflattened_priority_weights,
+                        cluster_spec, recurse_deeper=False,
+                    )
+                )
+
+        priority = np.array(
+            [1.0 / unflattened_priority_weights[jid] for jid in job_ids]
+        ).reshape((m, 1))
+        proportional_tputs = self._proportional.get_throughputs(
+            throughputs, index, cluster_spec
+        )
+        weights = throughputs * (priority / proportional_tputs) * sfa
+
+        x = self._nash_x(sfa[:, 0], self._num_workers[0]).reshape((m, 1))
+        rates = (weights * x).sum(axis=1)
+        rates_dict = {jid: rates[i] for i, jid in enumerate(job_ids)}
+        if not recurse_deeper:
+            return rates_dict
+
+        discounts = np.ones(m)
+        for i, jid in enumerate(job_ids):
+            for other, rate_without in rates_minus_job[i].items():
+                if rate_without > 0:
+                    discounts[i] *= rates_dict[other] / rate_without
+        discounted = np.clip(x * discounts.reshape((m, 1)), 0.0, 1.0)
+        return self.unflatten(discounted, index), discounts

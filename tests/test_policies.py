@@ -224,3 +224,29 @@ class TestFactory:
     def test_unknown_raises(self):
         with pytest.raises(ValueError):
             get_policy("nope")
+
+
+class TestStrategyProofPerf:
+    def test_symmetric_jobs_no_discount(self):
+        from shockwave_amd.policies import MaxMinFairnessStrategyProofPolicyWithPerf
+
+        p = MaxMinFairnessStrategyProofPolicyWithPerf()
+        alloc, discounts = p.get_allocation(
+            mk_tputs([2.0, 2.0]), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        # identical jobs impose identical externalities
+        assert discounts[0] == pytest.approx(discounts[1], rel=1e-6)
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(
+            alloc[JobIdPair(1)][WT], rel=1e-6
+        )
+
+    def test_nash_capacity_respected(self):
+        from shockwave_amd.policies import MaxMinFairnessStrategyProofPolicyWithPerf
+
+        p = MaxMinFairnessStrategyProofPolicyWithPerf()
+        sf = {JobIdPair(0): 2, JobIdPair(1): 1, JobIdPair(2): 1}
+        alloc, _ = p.get_allocation(
+            mk_tputs([1.0, 2.0, 3.0]), sf, mk_prio(3), {WT: 2}
+        )
+        used = sum(alloc[j][WT] * sf[j] for j in alloc)
+        assert used <= 2 + 1e-6
