@@ -171,34 +171,55 @@ class Dispatcher:
             )
         return proc.returncode, stdout
 
-    def _dispatch_job_helper(self, job, worker_id, round_id):
+    def _dispatch_jobs_helper(self, job_descriptions, worker_id, round_id):
+        """One RunJob request = one worker (GPU) for one round.  A packed
+        pair arrives as two job_descriptions and runs CO-RESIDENT on the
+        same GPU — HIP's hardware scheduler time-slices them (the
+        reference used CUDA MPS percentages for this).
+        """
         gpu_id = self._gpu_queue.get()
+        results = []
         try:
-            command = self._construct_command(job, gpu_id, worker_id)
-            self.launch_job(job, command, worker_id, round_id, gpu_id)
-            steps, duration, log = self._get_steps_and_execution_time(
-                job["job_id"], worker_id, round_id
-            )
+            threads = []
+            out = {}
+
+            def run_one(job):
+                command = self._construct_command(job, gpu_id, worker_id)
+                self.launch_job(job, command, worker_id, round_id, gpu_id)
+                out[job["job_id"]] = self._get_steps_and_execution_time(
+                    job["job_id"], worker_id, round_id
+                )
+
+            for job in job_descriptions:
+                t = threading.Thread(target=run_one, args=(job,))
+                t.start()
+                threads.append(t)
+            for t in threads:
+                t.join()
+            for job in job_descriptions:
+                steps, duration, log = out.get(job["job_id"], (0, 0.0, ""))
+                results.append((job["job_id"], steps, duration, log))
         finally:
             self._gpu_queue.put(gpu_id)
-        self._worker_rpc_client.notify_scheduler(
-            worker_id, [(job["job_id"], steps, duration, log)]
-        )
+        self._worker_rpc_client.notify_scheduler(worker_id, results)
 
     def dispatch_jobs(self, job_descriptions, worker_id, round_id):
-        for job in job_descriptions:
-            self._pool.submit(
-                self._safe_dispatch, job, worker_id, round_id
-            )
+        self._pool.submit(
+            self._safe_dispatch, job_descriptions, worker_id, round_id
+        )
 
-    def _safe_dispatch(self, job, worker_id, round_id):
+    def _safe_dispatch(self, job_descriptions, worker_id, round_id):
         try:
-            self._dispatch_job_helper(job, worker_id, round_id)
+            self._dispatch_jobs_helper(job_descriptions, worker_id, round_id)
         except Exception:
-            logger.exception("dispatch of job %s failed", job.get("job_id"))
+            logger.exception(
+                "dispatch of jobs %s failed",
+                [j.get("job_id") for j in job_descriptions],
+            )
             try:
                 self._worker_rpc_client.notify_scheduler(
-                    worker_id, [(job["job_id"], 0, 0.0, "")]
+                    worker_id,
+                    [(j["job_id"], 0, 0.0, "") for j in job_descriptions],
                 )
             except Exception:
                 logger.exception("failed to notify scheduler of failure")

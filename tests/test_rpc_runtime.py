@@ -290,7 +290,7 @@ class TestDispatcherProcess:
         }
         # bypass CLI arg appending by running helper directly: the fake
         # command ignores the appended args because of the trailing true
-        d._dispatch_job_helper(job, worker_id=2, round_id=0)
+        d._dispatch_jobs_helper([job], worker_id=2, round_id=0)
         assert notified["worker_id"] == 2
         job_id, steps, duration, log = notified["jobs"][0]
         assert (job_id, steps, duration) == (9, 17, 3.5)
@@ -318,7 +318,7 @@ class TestDispatcherProcess:
             "mode": "static",
         }
         t = threading.Thread(
-            target=d._safe_dispatch, args=(job, 0, 0), daemon=True
+            target=d._safe_dispatch, args=([job], 0, 0), daemon=True
         )
         t.start()
         deadline = time.time() + 5
@@ -330,3 +330,46 @@ class TestDispatcherProcess:
         t.join(timeout=15)
         assert not t.is_alive()
         assert time.time() - start < 12
+
+
+class TestPackedDispatch:
+    def test_pair_runs_coresident_one_done(self, tmp_path):
+        """Two job_descriptions in one RunJob share ONE GPU slot and are
+        reported in a single Done with both ids."""
+        from shockwave_amd.runtime.dispatcher import Dispatcher
+
+        notified = {}
+
+        class FakeClient:
+            def notify_scheduler(self, worker_id, jobs):
+                notified["jobs"] = jobs
+
+        ckpt_dir = tmp_path / "ckpts"
+
+        def mk(job_id, steps):
+            d = ckpt_dir / f"job_id={job_id}" / ".gavel" / "round=0"
+            script = (
+                "import os; d=r'%s'; os.makedirs(d, exist_ok=True); "
+                "open(os.path.join(d,'worker=0.log'),'w').write("
+                "'[2026-01-01 00:00:00] [PROGRESS] [STEPS] %d\\n"
+                "[2026-01-01 00:00:00] [PROGRESS] [DURATION] 1.0\\n')"
+            ) % (str(d), steps)
+            return {
+                "job_id": job_id,
+                "command": f"python3 -c \"{script}\" || true; true",
+                "working_directory": ".",
+                "needs_data_dir": False,
+                "num_steps_arg": "--steps",
+                "num_steps": 100,
+                "mode": "static",
+            }
+
+        d = Dispatcher(
+            round_duration=10, gpu_ids=[0], worker_rpc_client=FakeClient(),
+            sched_addr="127.0.0.1", sched_port=1, run_dir=str(tmp_path),
+            data_dir=None, checkpoint_dir=str(ckpt_dir),
+        )
+        d._dispatch_jobs_helper([mk(1, 11), mk(2, 22)], worker_id=0,
+                                round_id=0)
+        got = {jid: steps for jid, steps, _, _ in notified["jobs"]}
+        assert got == {1: 11, 2: 22}
