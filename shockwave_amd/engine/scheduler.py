@@ -553,17 +553,40 @@ class RoundScheduler:
                 self._allocation = self._compute_allocation()
                 self._need_to_update_allocation = False
 
+        # physically, micro-tasks still in flight have unaccounted time:
+        # credit each dispatched job its elapsed time since dispatch
+        # (reference :3644-3665)
+        elapsed_job_time, elapsed_worker_time = {}, {}
+        if not self._simulate:
+            for job_id in self._current_worker_assignments:
+                single = job_id.singletons()[0]
+                dispatch_time = self._per_job_latest_timestamps.get(single)
+                if dispatch_time is None:
+                    continue
+                dispatch_time = max(dispatch_time, self._last_reset_time)
+                elapsed = current_time - dispatch_time
+                wids = self._current_worker_assignments[job_id]
+                wt = self._worker_id_to_worker_type_mapping[wids[0]]
+                elapsed_job_time.setdefault(job_id, {}).setdefault(wt, 0.0)
+                elapsed_job_time[job_id][wt] += elapsed
+                elapsed_worker_time[wt] = (
+                    elapsed_worker_time.get(wt, 0.0) + elapsed
+                )
+
         fractions = {}
         for worker_type in self._worker_types:
             fractions[worker_type] = {}
             worker_time = self._worker_time_so_far[worker_type]
+            worker_time += elapsed_worker_time.get(worker_type, 0.0)
             for job_id in self._job_time_so_far:
                 if worker_time == 0.0 or worker_type not in self._job_time_so_far[job_id]:
                     fractions[worker_type][job_id] = 0.0
                 else:
-                    fractions[worker_type][job_id] = (
-                        self._job_time_so_far[job_id][worker_type] / worker_time
+                    job_time = self._job_time_so_far[job_id][worker_type]
+                    job_time += elapsed_job_time.get(job_id, {}).get(
+                        worker_type, 0.0
                     )
+                    fractions[worker_type][job_id] = job_time / worker_time
             for job_id in self._priorities[worker_type]:
                 if job_id not in self._allocation:
                     self._priorities[worker_type][job_id] = 0.0

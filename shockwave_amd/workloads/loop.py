@@ -168,10 +168,16 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
     # are enabled for static jobs only
     graphed = None
     static_batch = None
+    # capture costs seconds (warmup + MIOpen find on first process); only
+    # worth it when this lease will run enough steps to amortize it
+    expected_steps = target_steps - state["cumulative_steps"]
+    if lease_it is not None and lease_it._lease.max_steps < 1e9:
+        expected_steps = min(expected_steps, lease_it._lease.max_steps)
     if (
         device.type == "cuda"
         and mode == "static"
         and spec.make_static_batch is not None
+        and expected_steps >= int(os.environ.get("SWQ_GRAPH_MIN_STEPS", "100"))
         and os.environ.get("SWQ_GRAPHS", "1") != "0"
     ):
         from ..parallel.graphs import try_graph_step
