@@ -359,7 +359,12 @@ class PhysicalScheduler(RoundScheduler):
                 argument=(job_id,),
             )
             self._completion_events[job_id] = event
-        pool.submit(self._completion_event_scheduler.run)
+        try:
+            pool.submit(self._completion_event_scheduler.run)
+        except RuntimeError:
+            # pool already closed: only happens while shutting down
+            if not self._shutdown_event.is_set():
+                raise
 
     # ------------------------------------------------------------------
     # lease callbacks (reference :3880-4200)

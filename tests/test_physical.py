@@ -287,3 +287,73 @@ class TestWatchdog:
             assert len(sched.get_job_completion_times()) == 1
         finally:
             sched.shutdown()
+
+
+@pytest.mark.slow
+class TestMultiWorkerCluster:
+    def test_four_gpu_worker_mixed_jobs(self, tmp_path, throughputs):
+        """4 GPU slots on one node, 3 single-GPU jobs + 1 two-GPU job under
+        the shockwave planner: placement respects capacity, the DP job
+        spans two worker ids, and everything completes."""
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        port = free_port()
+        worker_port = free_port()
+        jobs = [tiny_job(steps=6) for _ in range(3)]
+        dp = Job(
+            job_id=None,
+            job_type="ResNet-18 (batch size 16)",
+            command="python3 main.py --batch_size 16",
+            working_directory="image_classification/cifar10",
+            num_steps_arg="--num_steps",
+            total_steps=8,
+            duration=600,
+            scale_factor=2,
+            mode="static",
+        )
+        jobs.append(dp)
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+        shockwave_config = {
+            "future_rounds": 4, "k": 1e-3, "lambda": 12.0, "rhomax": 1.0,
+            "time_per_iteration": 25, "num_gpus": 4,
+        }
+        sched = PhysicalScheduler(
+            get_policy("shockwave"),
+            port=port,
+            expected_num_workers=4,
+            throughputs=throughputs,
+            time_per_iteration=25,
+            profiles=profiles,
+            shockwave_config=shockwave_config,
+            worker_type="mi355x",
+        )
+        try:
+            Worker(
+                worker_type="mi355x",
+                sched_addr="127.0.0.1",
+                sched_port=port,
+                worker_port=worker_port,
+                num_gpus=4,
+                ip_addr="127.0.0.1",
+                run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                checkpoint_dir=str(tmp_path),
+            )
+            for j in jobs:
+                sched.add_job(j)
+            deadline = time.time() + 300
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(2)
+            completions = sched.get_job_completion_times()
+            assert len(completions) == 4, f"only {len(completions)}/4 done"
+            # capacity was never exceeded in any round
+            for rnd in sched.get_per_round_schedule():
+                assert sum(len(w) for w in rnd.values()) <= 4
+            # the DP job was placed on two distinct workers at least once
+            dp_id = jobs[-1].job_id[0]
+            dp_rounds = [r[dp_id] for r in sched.get_per_round_schedule()
+                         if dp_id in r]
+            assert any(len(set(w)) == 2 for w in dp_rounds)
+        finally:
+            sched.shutdown()
