@@ -200,3 +200,52 @@ class TestEstimatedThroughputPacking:
         assert makespan > 0
         assert len(sched.get_job_completion_times()) == 8
         assert len(sched._reference_job_map) > 0
+
+
+class TestAccordionSimTwin:
+    def test_bs_up_and_down_over_long_job(self, throughputs):
+        """The simulator twin scales an accordion ResNet-18 job up when it
+        leaves the critical regime and back down when it re-enters
+        (epochs 150-159)."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.core.job import Job
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        spe32 = 1563
+        job = Job(
+            job_id=None,
+            job_type="ResNet-18 (batch size 32)",
+            command="python3 main.py --data_dir=%s/cifar10 --batch_size 32",
+            working_directory="image_classification/cifar10",
+            num_steps_arg="--num_steps",
+            total_steps=spe32 * 170,   # spans the 150-159 re-entry window
+            duration=1e6,
+            scale_factor=1,
+            mode="accordion",
+        )
+        prof = trace_mod.build_job_profile(job, throughputs)
+        # short rounds: at MI355X bs-256 throughput a 120 s round covers
+        # ~50 epochs and can jump clean over the 10-epoch re-entry window
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=15,
+            profiles=[prof], worker_type="mi355x",
+        )
+        bs_history = []
+        orig_update = Job.update_bs
+
+        def record(self, new_bs):
+            bs_history.append(new_bs)
+            orig_update(self, new_bs)
+
+        Job.update_bs = record
+        try:
+            sched.simulate({"mi355x": 1}, [0.0], [job])
+        finally:
+            Job.update_bs = orig_update
+        assert 256 in bs_history, "never scaled up"
+        assert 32 in bs_history, "never scaled back down"
+        up_idx = bs_history.index(256)
+        down_idx = bs_history.index(32)
+        assert up_idx < down_idx

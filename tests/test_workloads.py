@@ -138,3 +138,40 @@ class TestGNSLoopIntegration:
             assert steps == 3
         finally:
             del os.environ["SWQ_DATASET_LEN"]
+
+
+class TestAccordionScaleDown:
+    def test_reentering_critical_regime_requests_small_bs(self, tmp_path):
+        """A job already scaled up (bs 256, original 32) that re-enters the
+        critical regime (epochs 150-159 for ResNet-18) reports small_bs."""
+        from tests.test_rpc_runtime import FakeLeaseClient
+        from shockwave_amd.parallel.ckpt_stream import CheckpointStore
+        from shockwave_amd.models import resnet18_cifar
+        from shockwave_amd.ops.optim import FusedSGD
+        import os
+        import torch
+
+        # fabricate the checkpoint a previous (scaled-up) lease would leave
+        model = resnet18_cifar()
+        opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9)
+        store = CheckpointStore(str(tmp_path))
+        store.save({
+            "model": model.state_dict(),
+            "optimizer": opt.state_dict(),
+            "epoch": 150,           # inside the 150-159 critical window
+            "cumulative_steps": 0,
+            "original_bs": 32,
+        })
+
+        os.environ["SWQ_DATASET_LEN"] = "512"  # 2 steps/epoch at bs 256
+        try:
+            client = FakeLeaseClient([(int(1e9), 1e9)])
+            families.cifar10_main(
+                ["--batch_size", "256", "--num_steps", "50",
+                 "--checkpoint_dir", str(tmp_path),
+                 "--enable_gavel_iterator", "--mode", "accordion"],
+                client=client,
+            )
+            assert client.rr_calls == [(False, True)]
+        finally:
+            del os.environ["SWQ_DATASET_LEN"]
