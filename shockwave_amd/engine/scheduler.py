@@ -361,6 +361,12 @@ class RoundScheduler:
         if timestamp is None:
             timestamp = current_timestamp
         self._per_job_start_timestamps[job_id] = timestamp
+        if self._worker_ids and job.scale_factor > len(self._worker_ids):
+            logger.warning(
+                "job %s requests %d GPUs but the cluster has %d: it can "
+                "never be scheduled and will be failed once the cluster "
+                "drains", job_id, job.scale_factor, len(self._worker_ids),
+            )
         logger.info("[Job dispatched] Job ID: %s duration: %s", job_id, job.duration)
         return job_id
 
@@ -1301,6 +1307,31 @@ class RoundScheduler:
                 # recompute so the clock can advance
                 self._need_to_update_allocation = True
                 self._last_reset_time = -self._minimum_time_between_allocation_resets
+                scheduled_jobs = self._schedule_jobs_on_workers()
+            if not scheduled_jobs and self._jobs and not queued_jobs:
+                # jobs larger than the cluster can never run: fail them
+                # rather than wedging the clock
+                total_gpus = len(self._worker_ids)
+                oversized = [
+                    jid for jid, job in self._jobs.items()
+                    if job.scale_factor > total_gpus
+                ]
+                for jid in oversized:
+                    logger.error(
+                        "job %s needs %d GPUs but the cluster has %d; "
+                        "marking failed", jid,
+                        self._jobs[jid].scale_factor, total_gpus,
+                    )
+                    self._per_job_latest_timestamps[jid] = (
+                        self.get_current_timestamp()
+                    )
+                    self._num_failures_per_job[jid] = MAX_FAILED_ATTEMPTS
+                    self._remove_job(jid)
+                    remaining_jobs -= 1
+                    if self.is_shockwave and jid[0] in self._shockwave_planner.metadata:
+                        self._shockwave_planner.remove_metadata(jid[0])
+                if not self._jobs:
+                    continue
                 scheduled_jobs = self._schedule_jobs_on_workers()
                 if not scheduled_jobs:
                     raise RuntimeError(

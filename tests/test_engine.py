@@ -249,3 +249,28 @@ class TestAccordionSimTwin:
         up_idx = bs_history.index(256)
         down_idx = bs_history.index(32)
         assert up_idx < down_idx
+
+
+class TestOversizedJob:
+    def test_job_larger_than_cluster_fails_not_wedges(self, throughputs):
+        """A job requesting more GPUs than exist is failed once the
+        cluster drains rather than deadlocking the clock (found by the
+        randomized sim sweep)."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+        from tests.test_core import make_job
+
+        small = make_job(steps=2000)
+        big = make_job(sf=8, steps=2000)
+        jobs = [small, big]
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        makespan = sched.simulate({"mi355x": 2}, [0.0, 0.0], jobs)
+        assert makespan > 0
+        completions = sched.get_job_completion_times()
+        assert len(completions) == 2  # small completed, big failed+recorded
