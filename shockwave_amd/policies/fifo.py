@@ -108,6 +108,11 @@ class FIFOPolicyWithPacking(PolicyWithPacking):
         for jid in queue:
             best, best_score = None, self._packing_threshold
             for sched in sorted(scheduled):
+                if sched.is_pair():
+                    # at most two jobs share a GPU; an existing pair is
+                    # not a packing candidate (and has no scale_factors
+                    # entry of its own)
+                    continue
                 if scale_factors[sched] != scale_factors[jid]:
                     continue
                 pair = JobIdPair(sched[0], jid[0])

@@ -250,3 +250,31 @@ class TestStrategyProofPerf:
         )
         used = sum(alloc[j][WT] * sf[j] for j in alloc)
         assert used <= 2 + 1e-6
+
+
+class TestFifoPacking:
+    def test_pair_not_repacked(self):
+        """A formed pair must not become a packing candidate itself.
+
+        Regression: with one GPU and three queued same-scale jobs whose
+        pairwise packed throughput clears the threshold, job 1 packs onto
+        job 0 forming a pair; job 2 then iterates the scheduled set which
+        contains that pair — looking up scale_factors[pair] raised
+        KeyError (pairs have no scale_factors entry), and Gavel packs at
+        most two jobs per GPU anyway.
+        """
+        from shockwave_amd.policies.fifo import FIFOPolicyWithPacking
+
+        p = FIFOPolicyWithPacking(packing_threshold=1.5)
+        singles = [JobIdPair(i) for i in range(3)]
+        tputs = {j: {WT: 1.0} for j in singles}
+        # every pair packs at 0.9+0.9 = 1.8 > 1.5 threshold
+        for a in range(3):
+            for b in range(a + 1, 3):
+                tputs[JobIdPair(a, b)] = {WT: (0.9, 0.9)}
+        sf = {j: 1 for j in singles}
+        alloc = p.get_allocation(tputs, sf, {WT: 1})
+        pair01 = JobIdPair(0, 1)
+        assert alloc[pair01][WT] == 1.0
+        # job 2 stays queued: the GPU already holds a full pair
+        assert all(v == 0.0 for v in alloc[JobIdPair(2)].values())
