@@ -6,20 +6,29 @@ from .fifo import FIFOPolicy, FIFOPolicyWithPerf, FIFOPolicyWithPacking
 from .max_min_fairness import (
     MaxMinFairnessPolicy,
     MaxMinFairnessPolicyWithPerf,
+    MaxMinFairnessPolicyWithPacking,
     MaxMinFairnessWaterFillingPolicy,
+    MaxMinFairnessWaterFillingPolicyWithPerf,
+    MaxMinFairnessWaterFillingPolicyWithPacking,
     MaxMinFairnessStrategyProofPolicy,
     MaxMinFairnessStrategyProofPolicyWithPerf,
 )
 from .finish_time_fairness import (
     FinishTimeFairnessPolicy,
     FinishTimeFairnessPolicyWithPerf,
+    FinishTimeFairnessPolicyWithPacking,
 )
 from .max_sum_throughput import (
     ThroughputSumWithPerf,
     ThroughputNormalizedByCostSumWithPerf,
     ThroughputNormalizedByCostSumWithPerfSLOs,
+    ThroughputNormalizedByCostSumWithPackingSLOs,
 )
-from .min_total_duration import MinTotalDurationPolicy, MinTotalDurationPolicyWithPerf
+from .min_total_duration import (
+    MinTotalDurationPolicy,
+    MinTotalDurationPolicyWithPerf,
+    MinTotalDurationPolicyWithPacking,
+)
 from .allox import AlloXPolicy
 from .gandiva import GandivaPolicy
 
@@ -34,27 +43,46 @@ class ShockwavePolicyStub(Policy):
 
 def get_policy(policy_name: str, seed=None, solver=None,
                priority_reweighting_policies=None):
+    # reference accepts "allox_alpha=<float>" spellings (utils.py:606-611)
+    if policy_name.startswith("allox"):
+        alpha = (
+            float(policy_name.split("allox_alpha=")[1])
+            if "alpha=" in policy_name
+            else 0.2
+        )
+        return AlloXPolicy(alpha=alpha)
     table = {
-        "allox": lambda: AlloXPolicy(),
         "fifo": lambda: FIFOPolicy(seed=seed),
         "fifo_perf": FIFOPolicyWithPerf,
         "fifo_packed": FIFOPolicyWithPacking,
         "finish_time_fairness": FinishTimeFairnessPolicy,
         "finish_time_fairness_perf": FinishTimeFairnessPolicyWithPerf,
+        "finish_time_fairness_packed": FinishTimeFairnessPolicyWithPacking,
         "gandiva": lambda: GandivaPolicy(seed=seed),
         "gandiva_fair": GandivaFairPolicy,
         "isolated": IsolatedPolicy,
         "isolated_plus": IsolatedPolicy,
         "max_min_fairness": MaxMinFairnessPolicy,
         "max_min_fairness_perf": MaxMinFairnessPolicyWithPerf,
-        "max_min_fairness_water_filling": MaxMinFairnessWaterFillingPolicy,
+        "max_min_fairness_packed": MaxMinFairnessPolicyWithPacking,
+        "max_min_fairness_water_filling": lambda: MaxMinFairnessWaterFillingPolicy(
+            priority_reweighting_policies=priority_reweighting_policies
+        ),
+        "max_min_fairness_water_filling_perf": lambda: MaxMinFairnessWaterFillingPolicyWithPerf(
+            priority_reweighting_policies=priority_reweighting_policies
+        ),
+        "max_min_fairness_water_filling_packed": lambda: MaxMinFairnessWaterFillingPolicyWithPacking(
+            priority_reweighting_policies=priority_reweighting_policies
+        ),
         "max_min_fairness_strategy_proof": MaxMinFairnessStrategyProofPolicy,
         "max_min_fairness_strategy_proof_perf": MaxMinFairnessStrategyProofPolicyWithPerf,
         "max_sum_throughput_perf": ThroughputSumWithPerf,
         "max_sum_throughput_normalized_by_cost_perf": ThroughputNormalizedByCostSumWithPerf,
         "max_sum_throughput_normalized_by_cost_perf_SLOs": ThroughputNormalizedByCostSumWithPerfSLOs,
+        "max_sum_throughput_normalized_by_cost_packed_SLOs": ThroughputNormalizedByCostSumWithPackingSLOs,
         "min_total_duration": MinTotalDurationPolicy,
         "min_total_duration_perf": MinTotalDurationPolicyWithPerf,
+        "min_total_duration_packed": MinTotalDurationPolicyWithPacking,
         "shockwave": ShockwavePolicyStub,
     }
     if policy_name not in table:

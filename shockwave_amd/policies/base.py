@@ -129,6 +129,7 @@ class PolicyWithPacking(Policy):
             {s for jid in job_ids for s in jid.singletons()}
         )
         m, n = len(job_ids), len(worker_types)
+        self._row_of = {jid: i for i, jid in enumerate(job_ids)}
         all_throughputs = []
         for single in single_job_ids:
             mat = np.zeros((m, n))
@@ -144,3 +145,50 @@ class PolicyWithPacking(Policy):
                         mat[i, j] = v
             all_throughputs.append(mat)
         return all_throughputs, (job_ids, worker_types), single_job_ids
+
+    def packed_constraints(
+        self,
+        m: int,
+        n: int,
+        scale_factors_array: np.ndarray,
+        job_ids,
+        single_job_ids,
+        extra_vars: int = 0,
+    ) -> Tuple[np.ndarray, np.ndarray]:
+        """A_ub, b_ub for the packed base constraints (policy.py:202-236):
+        per-worker-type capacity over ALL rows (singles and pairs), plus,
+        per SINGLE job, total time fraction across every combination row
+        involving it <= 1 (replaces the per-row constraint of
+        ``base_constraints`` — a job colocated in two pairs still only has
+        one unit of wall-clock)."""
+        nv = m * n + extra_vars
+        rows, rhs = [], []
+        for j in range(n):
+            row = np.zeros(nv)
+            for i in range(m):
+                row[i * n + j] = scale_factors_array[i, j]
+            rows.append(row)
+            rhs.append(self._num_workers[j])
+        for single in single_job_ids:
+            row = np.zeros(nv)
+            for i, jid in enumerate(job_ids):
+                if jid.overlaps_with(single):
+                    row[i * n : (i + 1) * n] = 1.0
+            rows.append(row)
+            rhs.append(1.0)
+        return np.array(rows), np.array(rhs)
+
+    def isolated_single_throughputs(self, all_throughputs, single_job_ids):
+        """(K x n) matrix of each single job's ISOLATED throughput — its
+        own singleton row of the packed tensor (every live job has one:
+        the engine registers singleton keys before any pair)."""
+        K = len(single_job_ids)
+        n = all_throughputs[0].shape[1]
+        iso = np.zeros((K, n))
+        for k, single in enumerate(single_job_ids):
+            i = self._row_of.get(single)
+            if i is not None:
+                iso[k] = all_throughputs[k][i]
+            else:  # defensive: best over combinations
+                iso[k] = np.max(all_throughputs[k], axis=0)
+        return iso

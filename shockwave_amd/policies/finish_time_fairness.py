@@ -23,7 +23,7 @@ import copy
 
 import numpy as np
 
-from .base import Policy
+from .base import Policy, PolicyWithPacking
 from .simple import IsolatedPolicy
 
 
@@ -167,3 +167,118 @@ class FinishTimeFairnessPolicy(Policy):
             num_steps_remaining,
             cluster_spec,
         )
+
+
+class FinishTimeFairnessPolicyWithPacking(PolicyWithPacking):
+    """Packed Themis: rho bisection where each SINGLE job's rate sums its
+    effective throughput over every combination row involving it
+    (reference finish_time_fairness.py:131-279).  Stateful bookkeeping
+    (cumulative isolated time) is keyed by single job, as in the perf
+    variant."""
+
+    name = "FinishTimeFairness_Packing"
+
+    def __init__(self):
+        self._isolated = IsolatedPolicy()
+        self._cumulative_isolated_time = {}
+        self._isolated_throughputs_prev = {}
+        self._num_steps_remaining_prev = {}
+
+    def _feasible(self, rho, m, n, w, rhs_steps, t_elapsed, t_iso,
+                  base_A, base_b):
+        rows, rhs = [], []
+        for k in range(len(rhs_steps)):
+            budget = rho * t_iso[k] - t_elapsed[k]
+            if budget <= 0:
+                if rhs_steps[k] > 0:
+                    return None
+                continue
+            rows.append(-w[k])
+            rhs.append(-rhs_steps[k] / budget)
+        if rows:
+            A = np.vstack([base_A, np.array(rows)])
+            b = np.concatenate([base_b, np.array(rhs)])
+        else:
+            A, b = base_A, base_b
+        res = self.solve_lp(np.zeros(m * n), A, b)
+        return res.x[: m * n].reshape((m, n)) if res.success else None
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        times_since_start,
+        num_steps_remaining,
+        cluster_spec,
+    ):
+        all_tputs, index, singles = self.flatten_packed(
+            unflattened_throughputs, cluster_spec
+        )
+        if all_tputs is None:
+            self._isolated_throughputs_prev = {}
+            self._num_steps_remaining_prev = {}
+            return None
+        job_ids, worker_types = index
+        m, n = all_tputs[0].shape
+        K = len(singles)
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+        w = np.array([t.reshape(-1) for t in all_tputs])
+
+        iso = self.isolated_single_throughputs(all_tputs, singles)
+        sf_singles = {s: scale_factors[s] for s in singles}
+        isolated_tputs = self._isolated.get_throughputs(
+            iso, (singles, worker_types), sf_singles, cluster_spec
+        ).reshape(-1)
+
+        t_elapsed = np.zeros(K)
+        t_iso = np.zeros(K)
+        rhs_steps = np.zeros(K)
+        for k, s in enumerate(singles):
+            self._cumulative_isolated_time.setdefault(s, 0.0)
+            if s in self._num_steps_remaining_prev:
+                steps_run = (
+                    self._num_steps_remaining_prev[s] - num_steps_remaining[s]
+                )
+                self._cumulative_isolated_time[s] += (
+                    steps_run / self._isolated_throughputs_prev[s]
+                )
+            t_elapsed[k] = times_since_start[s]
+            rhs_steps[k] = max(0.0, num_steps_remaining[s])
+            t_iso[k] = self._cumulative_isolated_time[s] + (
+                num_steps_remaining[s] / max(isolated_tputs[k], 1e-10)
+            )
+            t_iso[k] = max(t_iso[k], 1e-6)
+
+        base_A, base_b = self.packed_constraints(
+            m, n, sfa, job_ids, singles
+        )
+        lo, hi = 0.0, 2.0
+        x_best = None
+        for _ in range(60):
+            x = self._feasible(hi, m, n, w, rhs_steps, t_elapsed, t_iso,
+                               base_A, base_b)
+            if x is not None:
+                x_best = x
+                break
+            lo, hi = hi, hi * 2.0
+            if hi > 1e9:
+                break
+        if x_best is None:
+            return None
+        for _ in range(40):
+            mid = 0.5 * (lo + hi)
+            x = self._feasible(mid, m, n, w, rhs_steps, t_elapsed, t_iso,
+                               base_A, base_b)
+            if x is not None:
+                x_best, hi = x, mid
+            else:
+                lo = mid
+            if hi - lo <= 1e-3 * max(1.0, hi):
+                break
+
+        self._num_steps_remaining_prev = copy.copy(num_steps_remaining)
+        self._isolated_throughputs_prev = {
+            s: isolated_tputs[k] for k, s in enumerate(singles)
+        }
+        return self.unflatten(self.clip_allocation(x_best), index)

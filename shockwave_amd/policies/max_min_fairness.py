@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import numpy as np
 
-from .base import Policy
+from .base import Policy, PolicyWithPacking
 from .simple import ProportionalPolicy
 
 
@@ -91,13 +91,73 @@ class MaxMinFairnessPolicy(Policy):
         )
 
 
-class MaxMinFairnessWaterFillingPolicy(Policy):
+class MaxMinFairnessPolicyWithPacking(PolicyWithPacking):
+    """Packed max-min fairness: allocation rows span singles AND colocated
+    pairs; maximize the minimum normalized effective rate over SINGLE jobs,
+    where a single's rate sums its throughput contribution from every
+    combination row it appears in (reference max_min_fairness.py:220-410).
+    """
+
+    name = "MaxMinFairness_Packing"
+
+    def __init__(self):
+        self._proportional = ProportionalPolicy()
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+    ):
+        all_tputs, index, singles = self.flatten_packed(
+            unflattened_throughputs, cluster_spec
+        )
+        if all_tputs is None:
+            return None
+        job_ids, worker_types = index
+        m, n = all_tputs[0].shape
+        K = len(singles)
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+
+        iso = self.isolated_single_throughputs(all_tputs, singles)
+        prop = self._proportional.get_throughputs(
+            iso, (singles, worker_types), cluster_spec
+        ).reshape(-1)
+        prio = np.array(
+            [1.0 / unflattened_priority_weights[s] for s in singles]
+        )
+        sf_s = np.array([scale_factors[s] for s in singles], dtype=float)
+        scale = prio * sf_s / np.maximum(prop, 1e-10)
+
+        nv = m * n + 1
+        A_ub, b_ub = self.packed_constraints(
+            m, n, sfa, job_ids, singles, extra_vars=1
+        )
+        rows = []
+        for k in range(K):
+            row = np.zeros(nv)
+            row[: m * n] = -(all_tputs[k] * scale[k]).reshape(-1)
+            row[-1] = 1.0  # t - rate_k <= 0
+            rows.append(row)
+        A = np.vstack([A_ub, np.array(rows)])
+        b = np.concatenate([b_ub, np.zeros(K)])
+        c = np.zeros(nv)
+        c[-1] = -1.0
+        res = self.solve_lp(c, A, b)
+        if not res.success:
+            return None
+        x = self.clip_allocation(res.x[: m * n].reshape((m, n)))
+        return self.unflatten(x, index)
+
+
+class MaxMinFairnessWaterFillingPolicyWithPerf(Policy):
     """Water-filling max-min fairness: iteratively maximize the minimum
     normalized rate, freeze saturated jobs at the achieved level, recurse on
-    the rest (reference max_min_fairness_water_filling.py, 718 lines; this
+    the rest (reference max_min_fairness_water_filling.py:475-569; this
     is the standard algorithm, not a translation)."""
 
-    name = "MaxMinFairnessWaterFilling"
+    name = "MaxMinFairnessWaterFilling_Perf"
 
     def __init__(self, priority_reweighting_policies=None, max_iterations=64):
         self._max_iterations = max_iterations
@@ -186,6 +246,149 @@ class MaxMinFairnessWaterFillingPolicy(Policy):
                 for i in free:
                     frozen_level[i] = t_star
         return self.unflatten(self.clip_allocation(x_final), index)
+
+
+class MaxMinFairnessWaterFillingPolicy(Policy):
+    """LAS water filling: unit throughputs (time-fair), delegated to the
+    perf water filler (reference max_min_fairness_water_filling.py:416-473
+    substitutes throughput 1.0 for every (job, worker_type))."""
+
+    name = "MaxMinFairnessWaterFilling"
+
+    def __init__(self, priority_reweighting_policies=None, max_iterations=64):
+        self._perf = MaxMinFairnessWaterFillingPolicyWithPerf(
+            priority_reweighting_policies, max_iterations
+        )
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+        **kwargs,
+    ):
+        ones = {
+            jid: {wt: 1.0 for wt in per_wt}
+            for jid, per_wt in unflattened_throughputs.items()
+        }
+        return self._perf.get_allocation(
+            ones, scale_factors, unflattened_priority_weights, cluster_spec
+        )
+
+
+class MaxMinFairnessWaterFillingPolicyWithPacking(PolicyWithPacking):
+    """Water filling over packed rows: each round of filling maximizes the
+    minimum normalized rate over FREE single jobs (rates summed over every
+    combination row involving the single), freezes saturated singles at
+    the achieved level, and recurses (reference
+    max_min_fairness_water_filling.py:569-718)."""
+
+    name = "MaxMinFairnessWaterFilling_Packing"
+
+    def __init__(self, priority_reweighting_policies=None, max_iterations=64):
+        self._max_iterations = max_iterations
+        self._proportional = ProportionalPolicy()
+
+    def get_allocation(
+        self,
+        unflattened_throughputs,
+        scale_factors,
+        unflattened_priority_weights,
+        cluster_spec,
+        **kwargs,
+    ):
+        all_tputs, index, singles = self.flatten_packed(
+            unflattened_throughputs, cluster_spec
+        )
+        if all_tputs is None:
+            return None
+        job_ids, worker_types = index
+        m, n = all_tputs[0].shape
+        K = len(singles)
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+
+        iso = self.isolated_single_throughputs(all_tputs, singles)
+        prop = self._proportional.get_throughputs(
+            iso, (singles, worker_types), cluster_spec
+        ).reshape(-1)
+        prio = np.array(
+            [1.0 / unflattened_priority_weights[s] for s in singles]
+        )
+        sf_s = np.array([scale_factors[s] for s in singles], dtype=float)
+        scale = prio * sf_s / np.maximum(prop, 1e-10)
+        # per-single weight row over the flattened x
+        w = np.array(
+            [(all_tputs[k] * scale[k]).reshape(-1) for k in range(K)]
+        )
+
+        base_A, base_b = self.packed_constraints(
+            m, n, sfa, job_ids, singles, extra_vars=1
+        )
+        nv = m * n + 1
+        frozen_level = {}
+        x_final = np.zeros(m * n)
+        for _ in range(self._max_iterations):
+            free = [k for k in range(K) if k not in frozen_level]
+            if not free:
+                break
+            rows, rhs = [], []
+            for k in free:
+                row = np.zeros(nv)
+                row[: m * n] = -w[k]
+                row[-1] = 1.0
+                rows.append(row)
+                rhs.append(0.0)
+            for k, level in frozen_level.items():
+                row = np.zeros(nv)
+                row[: m * n] = -w[k]
+                rows.append(row)
+                rhs.append(-level)
+            A = np.vstack([base_A, np.array(rows)])
+            b = np.concatenate([base_b, np.array(rhs)])
+            c = np.zeros(nv)
+            c[-1] = -1.0
+            res = self.solve_lp(c, A, b)
+            if not res.success:
+                break
+            t_star = res.x[-1]
+            x_final = res.x[: m * n]
+            # freeze free singles whose rate cannot be raised above t_star
+            newly_frozen = False
+            for k in free:
+                c2 = np.zeros(nv)
+                c2[: m * n] = -w[k]
+                rows2, rhs2 = [], []
+                for o in free:
+                    if o == k:
+                        continue
+                    row = np.zeros(nv)
+                    row[: m * n] = -w[o]
+                    rows2.append(row)
+                    rhs2.append(-t_star + 1e-9)
+                for o, level in frozen_level.items():
+                    row = np.zeros(nv)
+                    row[: m * n] = -w[o]
+                    rows2.append(row)
+                    rhs2.append(-level)
+                A2 = (
+                    np.vstack([base_A, np.array(rows2)]) if rows2 else base_A
+                )
+                b2 = (
+                    np.concatenate([base_b, np.array(rhs2)])
+                    if rows2
+                    else base_b
+                )
+                res2 = self.solve_lp(c2, A2, b2)
+                best = -res2.fun if res2.success else t_star
+                if best <= t_star * (1 + 1e-6) + 1e-9:
+                    frozen_level[k] = t_star
+                    newly_frozen = True
+            if not newly_frozen:
+                for k in free:
+                    frozen_level[k] = t_star
+        x = self.clip_allocation(x_final.reshape((m, n)))
+        return self.unflatten(x, index)
 
 
 class MaxMinFairnessStrategyProofPolicy(MaxMinFairnessPolicy):

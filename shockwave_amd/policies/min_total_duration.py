@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import numpy as np
 
-from .base import Policy
+from .base import Policy, PolicyWithPacking
 
 
 class MinTotalDurationPolicyWithPerf(Policy):
@@ -85,3 +85,58 @@ class MinTotalDurationPolicy(Policy):
         return self._perf.get_allocation(
             new_tputs, scale_factors, num_steps_remaining, cluster_spec
         )
+
+
+class MinTotalDurationPolicyWithPacking(PolicyWithPacking):
+    """Packed OSSP: bisection on horizon T where the feasibility LP ranges
+    over singles AND pairs, and a single job's finish constraint sums its
+    effective throughput over every combination row involving it
+    (reference min_total_duration.py:135-234)."""
+
+    name = "MinTotalDuration_Packing"
+
+    def _feasible(self, T, m, n, w, steps, sfa, job_ids, singles):
+        A_ub, b_ub = self.packed_constraints(m, n, sfa, job_ids, singles)
+        rows = np.zeros((len(singles), m * n))
+        for k in range(len(singles)):
+            rows[k] = -w[k]
+        A = np.vstack([A_ub, rows])
+        b = np.concatenate([b_ub, -steps / T])
+        res = self.solve_lp(np.zeros(m * n), A, b)
+        return res.x[: m * n].reshape((m, n)) if res.success else None
+
+    def get_allocation(
+        self, unflattened_throughputs, scale_factors, num_steps_remaining,
+        cluster_spec
+    ):
+        all_tputs, index, singles = self.flatten_packed(
+            unflattened_throughputs, cluster_spec
+        )
+        if all_tputs is None:
+            return None
+        job_ids, worker_types = index
+        m, n = all_tputs[0].shape
+        sfa = self.scale_factors_array(scale_factors, job_ids, m, n)
+        w = np.array([t.reshape(-1) for t in all_tputs])
+        steps = np.array(
+            [max(0.0, num_steps_remaining[s]) for s in singles]
+        )
+
+        max_T, min_T = 1e6, 100.0
+        last_max_T = max_T
+        x_best = None
+        while x_best is None:
+            while 1.05 * min_T < max_T:
+                T = 0.5 * (min_T + max_T)
+                x = self._feasible(T, m, n, w, steps, sfa, job_ids, singles)
+                if x is not None:
+                    x_best, max_T = x, T
+                else:
+                    min_T = T
+            if x_best is not None:
+                break
+            min_T, max_T = last_max_T, last_max_T * 10.0
+            last_max_T *= 10.0
+            if last_max_T > 1e14:
+                return None
+        return self.unflatten(self.clip_allocation(x_best), index)

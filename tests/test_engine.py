@@ -149,7 +149,17 @@ class TestPackingPolicies:
     """Space-sharing simulation: pair throughputs populated and rounds
     scheduled without capacity violations."""
 
-    @pytest.mark.parametrize("policy", ["gandiva", "fifo_packed"])
+    @pytest.mark.parametrize(
+        "policy",
+        [
+            "gandiva",
+            "fifo_packed",
+            "max_min_fairness_packed",
+            "finish_time_fairness_packed",
+            "min_total_duration_packed",
+            "max_min_fairness_water_filling_packed",
+        ],
+    )
     def test_packing_sim_completes(self, policy, throughputs):
         r = sim(policy)
         assert len(r["jct_list"]) == 8

@@ -278,3 +278,131 @@ class TestFifoPacking:
         assert alloc[pair01][WT] == 1.0
         # job 2 stays queued: the GPU already holds a full pair
         assert all(v == 0.0 for v in alloc[JobIdPair(2)].values())
+
+
+def mk_packed(n, iso=2.0, pair=1.4):
+    """n singles, all pairs colocatable at per-job throughput ``pair``."""
+    tputs = {JobIdPair(i): {WT: iso} for i in range(n)}
+    for a in range(n):
+        for b in range(a + 1, n):
+            tputs[JobIdPair(a, b)] = {WT: (pair, pair)}
+    return tputs
+
+
+class TestPackedLPPolicies:
+    """The reference's *_packed LP policy family (utils.py:603-686 names),
+    over our packed flatten/constraint machinery (policy.py:87-236)."""
+
+    def test_max_min_fairness_packed_prefers_pairs(self):
+        from shockwave_amd.policies import MaxMinFairnessPolicyWithPacking
+
+        # 3 jobs on 1 GPU: time-slicing singles gives each rate 2/3;
+        # rotating pairs gives each 2*(1/3)*1.4 = 0.93 — LP must pack
+        alloc = MaxMinFairnessPolicyWithPacking().get_allocation(
+            mk_packed(3), mk_sf(3), mk_prio(3), {WT: 1}
+        )
+        pair_time = sum(
+            alloc[j][WT] for j in alloc if j.is_pair()
+        )
+        assert pair_time == pytest.approx(1.0, abs=1e-3)
+
+    def test_max_min_fairness_packed_skips_bad_pairs(self):
+        from shockwave_amd.policies import MaxMinFairnessPolicyWithPacking
+
+        # pair throughput 0.4 each (total 0.8 < 1.0 isolated-normalized):
+        # packing hurts, LP should time-slice singles instead
+        alloc = MaxMinFairnessPolicyWithPacking().get_allocation(
+            mk_packed(2, iso=2.0, pair=0.4), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        assert alloc[JobIdPair(0, 1)][WT] == pytest.approx(0.0, abs=1e-3)
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(0.5, abs=1e-2)
+
+    def test_min_total_duration_packed(self):
+        from shockwave_amd.policies import MinTotalDurationPolicyWithPacking
+
+        nsr = {JobIdPair(i): 1000.0 for i in range(3)}
+        alloc = MinTotalDurationPolicyWithPacking().get_allocation(
+            mk_packed(3), mk_sf(3), nsr, {WT: 1}
+        )
+        # total GPU time fraction across rows <= 1
+        used = sum(alloc[j][WT] for j in alloc)
+        assert used <= 1.0 + 1e-6
+        assert any(alloc[j][WT] > 0 for j in alloc if j.is_pair())
+
+    def test_finish_time_fairness_packed(self):
+        from shockwave_amd.policies import FinishTimeFairnessPolicyWithPacking
+
+        p = FinishTimeFairnessPolicyWithPacking()
+        tss = {JobIdPair(i): 100.0 for i in range(3)}
+        nsr = {JobIdPair(i): 500.0 for i in range(3)}
+        alloc = p.get_allocation(
+            mk_packed(3), mk_sf(3), mk_prio(3), tss, nsr, {WT: 1}
+        )
+        rates = []
+        for i in range(3):
+            r = alloc[JobIdPair(i)][WT] * 2.0
+            for j in alloc:
+                if j.is_pair() and j.overlaps_with(JobIdPair(i)):
+                    r += alloc[j][WT] * 1.4
+            rates.append(r)
+        # symmetric jobs -> symmetric packed rates
+        assert max(rates) == pytest.approx(min(rates), abs=0.05)
+
+    def test_mst_packed_slos(self):
+        from shockwave_amd.policies import (
+            ThroughputNormalizedByCostSumWithPackingSLOs,
+        )
+
+        p = ThroughputNormalizedByCostSumWithPackingSLOs()
+        alloc = p.get_allocation(mk_packed(2), mk_sf(2), {WT: 1})
+        # sum objective picks the pair (2.8 total > 2.0 single)
+        assert alloc[JobIdPair(0, 1)][WT] == pytest.approx(1.0, abs=1e-3)
+        # an SLO forcing job 0 to run alone beats the pair
+        alloc = p.get_allocation(
+            mk_packed(2), mk_sf(2), {WT: 1},
+            SLOs={JobIdPair(0): 100.0},
+            num_steps_remaining={JobIdPair(0): 190.0, JobIdPair(1): 190.0},
+        )
+        # job 0 needs rate 1.9 > 1.4 packed: solo time must cover the gap
+        rate0 = (
+            alloc[JobIdPair(0)][WT] * 2.0 + alloc[JobIdPair(0, 1)][WT] * 1.4
+        )
+        assert rate0 >= 1.9 - 1e-6
+
+    def test_water_filling_packed(self):
+        from shockwave_amd.policies import (
+            MaxMinFairnessWaterFillingPolicyWithPacking,
+        )
+
+        alloc = MaxMinFairnessWaterFillingPolicyWithPacking().get_allocation(
+            mk_packed(3), mk_sf(3), mk_prio(3), {WT: 1}
+        )
+        pair_time = sum(alloc[j][WT] for j in alloc if j.is_pair())
+        assert pair_time == pytest.approx(1.0, abs=1e-2)
+
+    def test_water_filling_base_is_las(self):
+        """Base water filling uses unit throughputs (reference
+        max_min_fairness_water_filling.py:443-447): a slow job gets the
+        same TIME share as a fast one."""
+        alloc = MaxMinFairnessWaterFillingPolicy().get_allocation(
+            mk_tputs([10.0, 1.0]), mk_sf(2), mk_prio(2), {WT: 1}
+        )
+        assert alloc[JobIdPair(0)][WT] == pytest.approx(
+            alloc[JobIdPair(1)][WT], abs=1e-3
+        )
+
+    def test_factory_packed_names(self):
+        for name in (
+            "max_min_fairness_packed",
+            "finish_time_fairness_packed",
+            "min_total_duration_packed",
+            "max_sum_throughput_normalized_by_cost_packed_SLOs",
+            "max_min_fairness_water_filling_perf",
+            "max_min_fairness_water_filling_packed",
+        ):
+            p = get_policy(name)
+            assert "Packing" in p.name or "Perf" in p.name
+
+    def test_allox_alpha_parsing(self):
+        assert get_policy("allox_alpha=0.7")._alpha == pytest.approx(0.7)
+        assert get_policy("allox")._alpha == pytest.approx(0.2)
