@@ -1190,7 +1190,7 @@ class RoundScheduler:
     def simulate(self, cluster_spec, arrival_times, jobs,
                  num_gpus_per_server=None, debug=False,
                  checkpoint_threshold=None, checkpoint_file=None,
-                 _resume_state=None):
+                 jobs_to_complete=None, _resume_state=None):
         if _resume_state is None:
             queued_jobs = list(zip(arrival_times, jobs))
             remaining_jobs = len(jobs)
@@ -1214,6 +1214,12 @@ class RoundScheduler:
 
         while True:
             if remaining_jobs == 0:
+                break
+            # windowed (steady-state) sims stop once the measurement-window
+            # jobs have all finished (reference simulate jobs_to_complete,
+            # scheduler.py:1728-1760)
+            if jobs_to_complete is not None and self.is_done(jobs_to_complete):
+                logger.info("all jobs in measurement window complete")
                 break
             next_job_arrival_time = queued_jobs[0][0] if queued_jobs else None
 

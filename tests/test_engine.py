@@ -284,3 +284,57 @@ class TestOversizedJob:
         assert makespan > 0
         completions = sched.get_job_completion_times()
         assert len(completions) == 2  # small completed, big failed+recorded
+
+
+class TestJobsToComplete:
+    def test_windowed_sim_stops_early(self, throughputs):
+        """simulate(jobs_to_complete=...) stops once the measurement
+        window completes (reference scheduler.py:1728-1760)."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        from shockwave_amd.core.job import JobIdPair
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        window = {JobIdPair(0), JobIdPair(1)}
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        sched.simulate({"mi355x": 2}, arrivals, jobs,
+                       jobs_to_complete=window)
+        completed = set(sched.get_job_completion_times().keys())
+        assert window.issubset(completed)
+        # early stop: the full 8-job trace was NOT run to completion
+        assert len(completed) < 8
+
+
+class TestSweepHarness:
+    def test_run_sweep_continuous(self, tmp_path):
+        """scripts/sweeps/run_sweep.py (reference run_sweep_continuous):
+        grid runs in a process pool, results land in jsonl."""
+        import json as _json
+        import subprocess
+
+        out = subprocess.run(
+            [sys.executable,
+             os.path.join(os.path.dirname(__file__), "..", "scripts",
+                          "sweeps", "run_sweep.py"),
+             "--mode", "continuous", "--policies", "max_min_fairness",
+             "-n", "2", "-a", "30", "-b", "60", "-s", "1", "-e", "6",
+             "--margin_jobs", "2", "-c", "4", "--seeds", "0", "-p", "2",
+             "--max_duration", "2000", "-l", str(tmp_path)],
+            capture_output=True, text=True, timeout=240,
+        )
+        assert out.returncode == 0, out.stderr[-2000:]
+        lines = [
+            _json.loads(line)
+            for line in open(tmp_path / "sweep_results.jsonl")
+        ]
+        assert len(lines) == 2
+        assert all(r["status"] == "ok" for r in lines)
+        assert all(r["avg_jct_s"] > 0 for r in lines)
