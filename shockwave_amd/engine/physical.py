@@ -46,6 +46,16 @@ from .scheduler import (
 
 logger = logging.getLogger("shockwave_amd.engine.physical")
 
+# Rounds an extended-lease job may go without a renewal before it is
+# declared dead and killed.  The reference kills after ONE silent round
+# (:4283-4339), safe at its 360 s rounds; at short rounds (tests, small
+# clusters) a renewal can land just past a round boundary — the 75%
+# renewal cadence gives ~0.75x round between renewals, plus process
+# startup — and killing on the first miss can livelock into a
+# kill/redispatch cycle that never accumulates steps.  Two silent rounds
+# cannot happen while a job is alive and renewing.
+EXTENDED_LEASE_GRACE_ROUNDS = 2
+
 
 class PhysicalScheduler(RoundScheduler):
     def __init__(
@@ -70,6 +80,7 @@ class PhysicalScheduler(RoundScheduler):
         self._current_round_start_time = None
         self._next_worker_assignments = None
         self._redispatched_worker_assignments = OrderedDict()
+        self._unresponsive_rounds = {}
         self._completed_jobs_in_current_round = set()
         self._completion_events = {}
         self._completion_event_scheduler = sched_module.scheduler(
@@ -570,12 +581,31 @@ class PhysicalScheduler(RoundScheduler):
                 for s in job_id.singletons()
             ]
             if min(num_updates) < scale_factor:
-                logger.error(
-                    "job %s held an extended lease but was unresponsive",
-                    job_id,
-                )
-                kill = True
+                misses = self._unresponsive_rounds.get(job_id, 0) + 1
+                self._unresponsive_rounds[job_id] = misses
+                if misses >= EXTENDED_LEASE_GRACE_ROUNDS:
+                    logger.error(
+                        "job %s held an extended lease but was unresponsive "
+                        "for %d consecutive rounds",
+                        job_id, misses,
+                    )
+                    kill = True
+                else:
+                    logger.warning(
+                        "job %s: no lease renewal this round (%d/%d before "
+                        "kill)",
+                        job_id, misses, EXTENDED_LEASE_GRACE_ROUNDS,
+                    )
+                    # tolerate this round: mark the job round-complete so
+                    # _end_round can proceed
+                    if job_id in self._completion_events:
+                        self._completed_jobs_in_current_round.add(job_id)
+                        del self._completion_events[job_id]
+                    for single in job_id.singletons():
+                        self._lease_update_requests[single] = []
+                        self._max_steps[single] = None
             elif job_id in self._completion_events:
+                self._unresponsive_rounds.pop(job_id, None)
                 self._completed_jobs_in_current_round.add(job_id)
                 del self._completion_events[job_id]
                 for single in job_id.singletons():
@@ -584,6 +614,7 @@ class PhysicalScheduler(RoundScheduler):
             if not kill:
                 self._scheduler_cv.notify_all()
         if kill:
+            self._unresponsive_rounds.pop(job_id, None)
             self._kill_job(job_id)
 
     # ------------------------------------------------------------------

@@ -161,16 +161,50 @@ class TestExtendedLeaseWatchdog:
         assert jid in sched._completed_jobs_in_current_round
         assert jid not in sched._completion_events
 
-    def test_unresponsive_job_killed(self, sched, throughputs):
-        """No lease renewals during the round -> declared unresponsive and
-        killed; with no live workers the kill degrades to a warning."""
+    def test_unresponsive_job_killed_after_grace(self, sched, throughputs):
+        """No lease renewals -> one grace round (the round still completes
+        for the job so _end_round can proceed), then killed on the second
+        consecutive silent round.  Killing on the FIRST miss livelocks
+        short-round configs: a renewal landing just past a round boundary
+        triggered kill -> redispatch -> slow start -> miss again."""
         jid, _ = add_job(sched, throughputs)
         sched._current_worker_assignments = OrderedDict({jid: (0,)})
         sched._jobs_with_extended_lease.add(jid)
         sched._lease_update_requests[jid] = []
+        sched._completion_events[jid] = object()
 
         killed = []
         sched._kill_job = lambda j: killed.append(j)
         sched._done_callback_extended_lease(jid)
+        assert killed == []  # grace round
+        assert jid in sched._completed_jobs_in_current_round
+
+        # second consecutive silent round -> kill
+        sched._completed_jobs_in_current_round = set()
+        sched._lease_update_requests[jid] = []
+        sched._done_callback_extended_lease(jid)
         assert killed == [jid]
         assert jid not in sched._completed_jobs_in_current_round
+
+    def test_renewal_resets_grace_counter(self, sched, throughputs):
+        """A renewal between two silent rounds resets the counter: only
+        CONSECUTIVE misses kill."""
+        jid, _ = add_job(sched, throughputs)
+        sched._current_worker_assignments = OrderedDict({jid: (0,)})
+        sched._jobs_with_extended_lease.add(jid)
+        killed = []
+        sched._kill_job = lambda j: killed.append(j)
+
+        sched._lease_update_requests[jid] = []
+        sched._completion_events[jid] = object()
+        sched._done_callback_extended_lease(jid)  # miss 1
+        # responsive round resets
+        sched._lease_update_requests[jid] = [(10, 5.0, 100, 60.0)]
+        sched._completion_events[jid] = object()
+        sched._done_callback_extended_lease(jid)
+        # next miss is miss 1 again, not 2
+        sched._completed_jobs_in_current_round = set()
+        sched._lease_update_requests[jid] = []
+        sched._completion_events[jid] = object()
+        sched._done_callback_extended_lease(jid)
+        assert killed == []
