@@ -1288,6 +1288,11 @@ class RoundScheduler:
                 queued_jobs.pop(0)
 
             if len(self._jobs) == 0:
+                if queued_jobs:
+                    # arrival gap: every active job finished before the next
+                    # arrival; jump the clock to it (next loop iteration
+                    # sets current_timestamp = next arrival and admits)
+                    continue
                 logger.warning("simulation complete: no jobs left")
                 break
 

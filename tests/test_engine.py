@@ -338,3 +338,28 @@ class TestSweepHarness:
         assert len(lines) == 2
         assert all(r["status"] == "ok" for r in lines)
         assert all(r["avg_jct_s"] > 0 for r in lines)
+
+
+class TestArrivalGap:
+    def test_sparse_arrivals_not_truncated(self, throughputs):
+        """All active jobs finishing before the next arrival must not end
+        the simulation: low-load (sparse interarrival) traces previously
+        dropped every job after the first gap."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, _ = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+        # arrivals 10x sparser than any job's duration: guaranteed gaps
+        arrivals = [i * 500000.0 for i in range(len(jobs))]
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        sched.simulate({"mi355x": 2}, arrivals, jobs)
+        assert len(sched.get_job_completion_times()) == 8
