@@ -88,6 +88,17 @@ class PhysicalScheduler(RoundScheduler):
         )
         self._shutdown_event = threading.Event()
 
+        # hang diagnosis: periodic all-thread stack dumps (reference
+        # faulthandler.dump_traceback_later hook, scheduler.py:450-455)
+        from ..utils.logging import enable_hang_diagnosis
+
+        try:
+            self._cancel_hang_diagnosis = enable_hang_diagnosis(
+                ".stack_trace.log", interval_s=30.0
+            )
+        except OSError:
+            self._cancel_hang_diagnosis = lambda: None
+
         self._server = serve_scheduler(
             port,
             {
@@ -811,3 +822,7 @@ class PhysicalScheduler(RoundScheduler):
                         pass
                     seen.add(key)
         self._server.stop(5)
+        try:
+            self._cancel_hang_diagnosis()
+        except Exception:
+            pass

@@ -1310,6 +1310,14 @@ class RoundScheduler:
                 )
                 checkpoint_saved = True
 
+            if debug:
+                # single-step rounds for interactive diagnosis (reference
+                # :1881-1882)
+                input(
+                    "t=%.1f jobs=%d> " % (self._current_timestamp,
+                                          len(self._jobs))
+                )
+
             # schedule the round
             scheduled_jobs = self._schedule_jobs_on_workers()
             if not scheduled_jobs and self._jobs and not queued_jobs:

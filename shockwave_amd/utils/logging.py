@@ -32,3 +32,25 @@ def build_logger(name: str, level=logging.INFO, log_file=None):
         fh.setFormatter(fmt)
         logger.addHandler(fh)
     return logger
+
+
+def enable_hang_diagnosis(path: str = ".stack_trace.log",
+                          interval_s: float = 30.0):
+    """Periodically dump every thread's stack to ``path`` — if the head
+    process wedges (lost cv notify, stuck RPC), the last dump shows where
+    (reference scheduler.py:450-455 faulthandler hook).
+
+    Returns a cancel() callable; the file handle stays open for
+    faulthandler's lifetime."""
+    import faulthandler
+
+    f = open(path, "w")
+    faulthandler.dump_traceback_later(
+        interval_s, repeat=True, file=f
+    )
+
+    def cancel():
+        faulthandler.cancel_dump_traceback_later()
+        f.close()
+
+    return cancel

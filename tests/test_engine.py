@@ -363,3 +363,42 @@ class TestArrivalGap:
         )
         sched.simulate({"mi355x": 2}, arrivals, jobs)
         assert len(sched.get_job_completion_times()) == 8
+
+
+class TestDebugAndDiagnosisHooks:
+    def test_debug_single_steps_rounds(self, throughputs, monkeypatch):
+        """simulate(debug=True) pauses at each round via input()
+        (reference scheduler.py:1881-1882)."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        prompts = []
+        monkeypatch.setattr(
+            "builtins.input", lambda p="": prompts.append(p) or ""
+        )
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        sched.simulate({"mi355x": 2}, arrivals, jobs, debug=True)
+        assert len(prompts) > 5
+        assert len(sched.get_job_completion_times()) == 8
+
+    def test_hang_diagnosis_dumps_stacks(self, tmp_path):
+        """enable_hang_diagnosis writes periodic all-thread stack dumps
+        (reference faulthandler hook, scheduler.py:450-455)."""
+        import time as _time
+
+        from shockwave_amd.utils.logging import enable_hang_diagnosis
+
+        path = str(tmp_path / "stacks.log")
+        cancel = enable_hang_diagnosis(path, interval_s=0.2)
+        _time.sleep(0.6)
+        cancel()
+        data = open(path).read()
+        assert "File" in data and "Thread" in data
