@@ -41,3 +41,31 @@ class RecommendationAutoencoder(nn.Module):
         """Masked MSE over observed interactions + negative sampling over
         the rest (simplified Recoder loss)."""
         return F.mse_loss(recon, target)
+
+
+def recall_at_k(scores: torch.Tensor, targets: torch.Tensor,
+                k: int = 20) -> torch.Tensor:
+    """Per-user Recall@k: hits in the top-k / min(#relevant, k)
+    (reference workloads/pytorch/recommendation recoder/metrics.py)."""
+    topk = scores.topk(k, dim=1).indices
+    hits = targets.gather(1, topk).sum(1)
+    denom = targets.sum(1).clamp(max=float(k)).clamp(min=1.0)
+    return hits / denom
+
+
+def ndcg_at_k(scores: torch.Tensor, targets: torch.Tensor,
+              k: int = 20) -> torch.Tensor:
+    """Per-user NDCG@k with binary relevance."""
+    topk = scores.topk(k, dim=1).indices
+    gains = targets.gather(1, topk)
+    discounts = 1.0 / torch.log2(
+        torch.arange(2, k + 2, dtype=scores.dtype, device=scores.device)
+    )
+    dcg = (gains * discounts).sum(1)
+    ideal_counts = targets.sum(1).clamp(max=float(k)).long()
+    cum = torch.cat(
+        [torch.zeros(1, dtype=scores.dtype, device=scores.device),
+         discounts.cumsum(0)]
+    )
+    idcg = cum[ideal_counts]
+    return dcg / idcg.clamp(min=1e-8)

@@ -175,3 +175,92 @@ class TestAccordionScaleDown:
             assert client.rr_calls == [(False, True)]
         finally:
             del os.environ["SWQ_DATASET_LEN"]
+
+
+class TestRecommendationUnit:
+    """Model-level unit tests for the recommendation family — the one
+    workload where the reference carries a real pytest suite
+    (workloads/pytorch/recommendation/tests/)."""
+
+    def _model(self, n=50):
+        from shockwave_amd.models.recommendation import (
+            RecommendationAutoencoder,
+        )
+
+        return RecommendationAutoencoder(num_items=n, hidden=(16, 8))
+
+    def test_forward_shape_roundtrip(self):
+        import torch
+
+        m = self._model(50).eval()
+        x = torch.rand(4, 50)
+        out = m(x)
+        assert out.shape == (4, 50)
+
+    def test_denoising_only_in_training(self):
+        import torch
+
+        torch.manual_seed(0)
+        m = self._model(50)
+        x = torch.rand(8, 50)
+        m.eval()
+        a, b = m(x), m(x)
+        assert torch.equal(a, b)  # no dropout noise in eval
+
+    def test_overfits_tiny_batch(self):
+        import torch
+
+        torch.manual_seed(0)
+        m = self._model(30)
+        m.noise_prob = 0.0
+        m.train()
+        x = (torch.rand(4, 30) > 0.8).float()
+        opt = torch.optim.Adam(m.parameters(), lr=1e-2)
+        first = None
+        for _ in range(120):
+            opt.zero_grad()
+            loss = m.loss(m(x), x)
+            loss.backward()
+            opt.step()
+            first = first if first is not None else loss.item()
+        assert loss.item() < first * 0.5
+
+    def test_recall_at_k_perfect_and_empty(self):
+        import torch
+
+        from shockwave_amd.models.recommendation import recall_at_k
+
+        targets = torch.zeros(2, 10)
+        targets[0, [1, 3]] = 1.0
+        scores = torch.full((2, 10), -1.0)
+        scores[0, 1], scores[0, 3] = 5.0, 4.0
+        r = recall_at_k(scores, targets, k=2)
+        assert r[0].item() == pytest.approx(1.0)
+        assert r[1].item() == pytest.approx(0.0)  # no relevant items
+
+    def test_recall_counts_partial_hits(self):
+        import torch
+
+        from shockwave_amd.models.recommendation import recall_at_k
+
+        targets = torch.zeros(1, 10)
+        targets[0, [0, 1, 2, 3]] = 1.0
+        scores = torch.arange(10, 0, -1).float().unsqueeze(0)
+        # top-2 = items 0,1 -> 2 hits / min(4 relevant, 2) = 1.0
+        assert recall_at_k(scores, targets, k=2)[0].item() == pytest.approx(1.0)
+        # top-8 = items 0..7 -> 4 hits / min(4, 8) = 1.0
+        assert recall_at_k(scores, targets, k=8)[0].item() == pytest.approx(1.0)
+
+    def test_ndcg_ordering_sensitivity(self):
+        import torch
+
+        from shockwave_amd.models.recommendation import ndcg_at_k
+
+        targets = torch.zeros(1, 6)
+        targets[0, 0] = 1.0
+        best = torch.tensor([[6.0, 5, 4, 3, 2, 1]])
+        worst_in_k = torch.tensor([[1.0, 6, 5, 4, 3, 2]])
+        n_best = ndcg_at_k(best, targets, k=3)[0].item()
+        n_late = ndcg_at_k(worst_in_k, targets, k=6)[0].item()
+        assert n_best == pytest.approx(1.0)
+        assert 0 < n_late < n_best
