@@ -758,6 +758,7 @@ class PhysicalScheduler(RoundScheduler):
                     self._worker_time_so_far[worker_type] += max_exec
                 for wid in all_worker_ids:
                     self._cumulative_worker_time_so_far[wid] += max_exec
+                self._accrue_cost(job_id, worker_type, max_exec)
 
             self._update_throughput(
                 job_id, worker_type, agg_steps, agg_times

@@ -73,6 +73,7 @@ class RoundScheduler:
         estimate_throughputs: bool = False,
         profiling_percentage: float = 1.0,
         num_reference_models: int = 16,
+        per_worker_type_prices: Optional[Dict[str, float]] = None,
     ):
         self._policy = policy
         self._simulate = simulate
@@ -123,6 +124,8 @@ class RoundScheduler:
         self._total_steps_run: Dict = {}
         self._job_time_so_far: Dict = {}
         self._job_cost_so_far: Dict = {}
+        self._per_worker_type_prices = per_worker_type_prices
+        self._SLOs: Dict = {}
         self._cumulative_run_time: Dict = {}
         self._num_failures_per_job: Dict = {}
         self._per_job_start_timestamps: Dict = {}
@@ -304,6 +307,8 @@ class RoundScheduler:
         self._job_id_counter += 1
         job.job_id = job_id
         self._jobs[job_id] = job
+        if job.SLO is not None:
+            self._SLOs[job_id] = job.SLO
         self._steps_run_so_far[job_id] = {}
         self._job_time_so_far[job_id] = {}
         self._job_cost_so_far[job_id] = 0.0
@@ -529,8 +534,20 @@ class RoundScheduler:
         elif name.startswith("ThroughputNormalizedByCostSum") or name.startswith(
             "ThroughputSum"
         ):
+            kwargs = {}
+            if name.startswith("ThroughputNormalizedByCostSum"):
+                kwargs["instance_costs"] = self._per_worker_type_prices or {
+                    wt: 1.0 for wt in self._worker_types
+                }
+            if "SLOs" in name:
+                kwargs["SLOs"] = {
+                    jid: self._SLOs[jid]
+                    for jid in throughputs
+                    if jid in self._SLOs
+                }
+                kwargs["num_steps_remaining"] = state["num_steps_remaining"]
             allocation = self._policy.get_allocation(
-                throughputs, scale_factors, cluster_spec
+                throughputs, scale_factors, cluster_spec, **kwargs
             )
         elif name == "Proportional":
             allocation = self._policy.get_allocation(throughputs, cluster_spec)
@@ -1117,6 +1134,7 @@ class RoundScheduler:
                 self._worker_time_so_far[worker_type] += max_exec
             for wid in all_worker_ids:
                 self._cumulative_worker_time_so_far[wid] += max_exec
+            self._accrue_cost(job_id, worker_type, max_exec)
 
         self._update_throughput(
             job_id, worker_type, agg_num_steps, agg_execution_times
@@ -1438,6 +1456,35 @@ class RoundScheduler:
     # ------------------------------------------------------------------
     # Metrics (reference :2779-3107)
     # ------------------------------------------------------------------
+
+    def _accrue_cost(self, job_id, worker_type, execution_time):
+        """Dollar cost of one micro-task at the per-hour worker price
+        (reference scheduler.py:4593-4604)."""
+        if self._per_worker_type_prices is None:
+            return
+        price = self._per_worker_type_prices.get(worker_type, 0.0)
+        for single in job_id.singletons():
+            job = self._jobs.get(single)
+            if job is None:
+                continue
+            self._job_cost_so_far[single] = (
+                self._job_cost_so_far.get(single, 0.0)
+                + price * execution_time / 3600.0 * job.scale_factor
+            )
+
+    def get_total_cost(self):
+        """Reference scheduler.py:3060-3066."""
+        return float(sum(self._job_cost_so_far.values()))
+
+    def get_num_SLO_violations(self):
+        """Jobs whose completion time exceeded their SLO (reference
+        scheduler.py:3068-3084)."""
+        violations = 0
+        for job_id, slo in self._SLOs.items():
+            ct = self._job_completion_times.get(job_id)
+            if ct is not None and ct > slo:
+                violations += 1
+        return violations
 
     def get_average_jct(self, job_ids=None):
         if not self._job_completion_times:

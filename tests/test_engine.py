@@ -402,3 +402,61 @@ class TestDebugAndDiagnosisHooks:
         cancel()
         data = open(path).read()
         assert "File" in data and "Thread" in data
+
+
+class TestCostAndSLOMetrics:
+    def _run(self, throughputs, policy="max_min_fairness", prices=None,
+             slo=None):
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+            if slo is not None:
+                j.SLO = slo
+        sched = RoundScheduler(
+            get_policy(policy), simulate=True, throughputs=throughputs,
+            time_per_iteration=120, profiles=profiles,
+            worker_type="mi355x", per_worker_type_prices=prices,
+        )
+        sched.simulate({"mi355x": 2}, arrivals, jobs)
+        return sched
+
+    def test_cost_accrues_at_worker_price(self, throughputs):
+        """Total $ = GPU-hours x price (reference scheduler.py:4593-4604,
+        3060-3066)."""
+        price = 12.0
+        sched = self._run(throughputs, prices={"mi355x": price})
+        total = sched.get_total_cost()
+        gpu_hours = (
+            sum(sched._cumulative_worker_time_so_far.values()) / 3600.0
+        )
+        assert total > 0
+        assert total == pytest.approx(gpu_hours * price, rel=0.05)
+
+    def test_no_prices_no_cost(self, throughputs):
+        sched = self._run(throughputs)
+        assert sched.get_total_cost() == 0.0
+
+    def test_slo_violations_counted(self, throughputs):
+        """Impossible 1s SLOs -> every completed job violates (reference
+        scheduler.py:3068-3084)."""
+        sched = self._run(throughputs, slo=1.0)
+        assert sched.get_num_SLO_violations() == 8
+        sched = self._run(throughputs, slo=1e9)
+        assert sched.get_num_SLO_violations() == 0
+
+    def test_slo_policy_sim_completes(self, throughputs):
+        """The SLO-constrained MST policy receives SLOs/steps/costs from
+        the engine and the sim completes."""
+        sched = self._run(
+            throughputs,
+            policy="max_sum_throughput_normalized_by_cost_perf_SLOs",
+            prices={"mi355x": 10.0},
+            slo=36000.0,
+        )
+        assert len(sched.get_job_completion_times()) == 8
