@@ -37,6 +37,14 @@ def run_simulation(
     import logging
 
     logging.basicConfig(level=getattr(logging, log_level))
+    if results_dir:
+        # mirror console output per policy (reference scheduler.py:126-133)
+        os.makedirs(results_dir, exist_ok=True)
+        fh = logging.FileHandler(
+            os.path.join(results_dir, f"console_output_{policy_name}.txt")
+        )
+        fh.setLevel(getattr(logging, log_level))
+        logging.getLogger().addHandler(fh)
 
     throughputs = read_throughputs(throughputs_file)
     worker_type = trace_mod.canonical_worker_type(throughputs)
