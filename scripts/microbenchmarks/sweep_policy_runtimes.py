@@ -35,6 +35,20 @@ def bench_policy(name, njobs, seed=0):
     cluster = {WT: max(4, njobs // 4)}
 
     policy = get_policy(name, seed=seed)
+    from shockwave_amd.policies import PolicyWithPacking
+
+    if isinstance(policy, PolicyWithPacking):
+        # packed LPs range over singles AND same-scale pairs, like the
+        # engine's pair registration — include them so the measured LP
+        # size is the real one
+        singles = list(tputs.keys())
+        for a in range(njobs):
+            for b in range(a + 1, njobs):
+                ja, jb = singles[a], singles[b]
+                if sf[ja] != sf[jb]:
+                    continue
+                t1, t2 = tputs[ja][WT] * 0.7, tputs[jb][WT] * 0.7
+                tputs[JobIdPair(ja[0], jb[0])] = {WT: (t1, t2)}
     t0 = time.time()
     if name == "allox":
         policy.get_allocation(tputs, sf, times, steps, [], cluster)
