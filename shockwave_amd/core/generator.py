@@ -90,6 +90,8 @@ def generate_job(
     mode: str = "static",
     run_dir: Optional[str] = None,
     job_table=None,
+    multi_priority: bool = False,
+    slo_rng: Optional[random.Random] = None,
 ) -> Job:
     if job_table is None:
         job_table = JobTable
@@ -117,6 +119,19 @@ def generate_job(
     num_steps = duration_s * throughputs[reference_worker_type][key]["null"]
     assert num_steps > 0
 
+    # optional priority tier: 20% of jobs at weight 5 (reference
+    # utils.py:242-247)
+    priority_weight = 1.0
+    if multi_priority and rng.uniform(0, 1) <= 0.2:
+        priority_weight = 5.0
+    # optional SLO: factor x ideal duration from {1.2, 2, 10}
+    # (reference utils.py:249-258)
+    SLO = -1.0
+    if slo_rng is not None:
+        r = slo_rng.uniform(0, 1)
+        factor = 1.2 if r < 0.33 else (2.0 if r < 0.67 else 10.0)
+        SLO = factor * duration_s
+
     return Job(
         job_id=None,
         job_type=template.model,
@@ -127,8 +142,8 @@ def generate_job(
         duration=duration_s,
         scale_factor=scale_factor,
         mode=mode,
-        priority_weight=1.0,
-        SLO=-1.0,
+        priority_weight=priority_weight,
+        SLO=SLO,
         needs_data_dir=template.needs_data_dir,
     )
 
@@ -147,6 +162,8 @@ def generate_trace(
     multi_gpu: bool = True,
     run_dir: Optional[str] = None,
     job_table=None,
+    multi_priority: bool = False,
+    generate_slos: bool = False,
 ) -> Tuple[List[Job], List[float]]:
     """Generate a TACC-style trace: returns (jobs, arrival_times)."""
     job_rng = random.Random(seed)
@@ -154,6 +171,7 @@ def generate_trace(
     duration_rng = random.Random(seed + 2)
     sf_rng = random.Random(seed + 3)
     mode_rng = random.Random(seed + 4)
+    slo_rng = random.Random(seed + 5) if generate_slos else None
 
     durations = duration_space(min_duration_s, max_duration_s, num_durations)
 
@@ -172,6 +190,8 @@ def generate_trace(
             mode=mode,
             run_dir=run_dir,
             job_table=job_table,
+            multi_priority=multi_priority,
+            slo_rng=slo_rng,
         )
         jobs.append(job)
         arrivals.append(round(t))

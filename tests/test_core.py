@@ -199,3 +199,38 @@ class TestGenerator:
         jobs2, arrivals2 = trace.parse_trace(str(p))
         assert len(jobs2) == 10
         assert [j.job_type for j in jobs2] == [j.job_type for j in jobs]
+
+
+class TestGeneratorPrioritySLO:
+    def test_multi_priority_and_slos(self, throughputs):
+        """reference utils.py:242-258: ~20% priority-5 jobs, SLO =
+        {1.2, 2, 10} x ideal duration."""
+        from shockwave_amd.core import generator
+
+        jobs, _ = generator.generate_trace(
+            throughputs, "mi355x", 80, seed=0,
+            multi_priority=True, generate_slos=True,
+        )
+        frac5 = sum(1 for j in jobs if j.priority_weight == 5.0) / len(jobs)
+        assert 0.05 < frac5 < 0.4
+        factors = {round(j.SLO / j.duration, 1) for j in jobs}
+        assert factors <= {1.2, 2.0, 10.0}
+        # defaults unchanged: no priorities/SLOs unless requested
+        jobs, _ = generator.generate_trace(throughputs, "mi355x", 10, seed=0)
+        assert all(j.priority_weight == 1.0 for j in jobs)
+        assert all(j.SLO is None for j in jobs)
+
+    def test_priority_respected_by_policy(self, throughputs):
+        """A priority-5 job gets a larger LAS share."""
+        import copy
+
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.core.job import JobIdPair
+        from shockwave_amd.policies import get_policy
+
+        p = get_policy("max_min_fairness")
+        tputs = {JobIdPair(0): {"mi355x": 1.0}, JobIdPair(1): {"mi355x": 1.0}}
+        sf = {JobIdPair(0): 1, JobIdPair(1): 1}
+        prio = {JobIdPair(0): 5.0, JobIdPair(1): 1.0}
+        alloc = p.get_allocation(tputs, sf, prio, {"mi355x": 1})
+        assert alloc[JobIdPair(0)]["mi355x"] > alloc[JobIdPair(1)]["mi355x"]
