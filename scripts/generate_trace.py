@@ -26,6 +26,10 @@ def main():
                    help="static,accordion,gns fractions")
     p.add_argument("--scale_factor_mix", default="0.6,0.3,0.09,0.01")
     p.add_argument("--single_gpu", action="store_true")
+    p.add_argument("--generate_multi_priority_jobs", action="store_true",
+                   help="~20%% of jobs get priority weight 5")
+    p.add_argument("--generate_SLOs", action="store_true",
+                   help="assign SLO = {1.2,2,10} x ideal duration")
     p.add_argument("-o", "--output_file", required=True)
     args = p.parse_args()
 
@@ -43,6 +47,8 @@ def main():
         mode_mix=[float(x) for x in args.mode_mix.split(",")],
         seed=args.seed,
         multi_gpu=not args.single_gpu,
+        multi_priority=args.generate_multi_priority_jobs,
+        generate_slos=args.generate_SLOs,
     )
     trace.write_trace(jobs, arrivals, args.output_file)
     modes = [j.mode for j in jobs]
