@@ -1,7 +1,12 @@
 """Policy registry (reference: utils.get_policy, utils.py:603-686)."""
 
 from .base import Policy, PolicyWithPacking
-from .simple import IsolatedPolicy, ProportionalPolicy, GandivaFairPolicy
+from .simple import (
+    IsolatedPolicy,
+    IsolatedPlusPolicy,
+    ProportionalPolicy,
+    GandivaFairPolicy,
+)
 from .fifo import FIFOPolicy, FIFOPolicyWithPerf, FIFOPolicyWithPacking
 from .max_min_fairness import (
     MaxMinFairnessPolicy,
@@ -61,7 +66,7 @@ def get_policy(policy_name: str, seed=None, solver=None,
         "gandiva": lambda: GandivaPolicy(seed=seed),
         "gandiva_fair": GandivaFairPolicy,
         "isolated": IsolatedPolicy,
-        "isolated_plus": IsolatedPolicy,
+        "isolated_plus": IsolatedPlusPolicy,
         "max_min_fairness": MaxMinFairnessPolicy,
         "max_min_fairness_perf": MaxMinFairnessPolicyWithPerf,
         "max_min_fairness_packed": MaxMinFairnessPolicyWithPacking,

@@ -47,6 +47,24 @@ class IsolatedPolicy(Policy):
         return self.unflatten(x, index)
 
 
+class IsolatedPlusPolicy(IsolatedPolicy):
+    """Like Isolated but a job's share is NOT divided by its scale
+    factor — an sf-GPU job receives a full equal share per GPU
+    (isolated_plus.py:35-56; the only diff from isolated.py is the
+    removed scale-factor division at :50)."""
+
+    name = "Isolated_Plus"
+
+    def _allocation(self, m, n, scale_factors_array, worker_types,
+                    cluster_spec):
+        x = np.array(
+            [[cluster_spec[wt] / m for wt in worker_types] for _ in range(m)],
+            dtype=float,
+        )
+        row_sum = np.maximum(x.sum(axis=1), 1.0)
+        return x / row_sum[:, None]
+
+
 class ProportionalPolicy(Policy):
     """Cluster split proportional to worker counts, normalized by the max
     row sum (proportional.py:26-43)."""

@@ -406,3 +406,21 @@ class TestPackedLPPolicies:
     def test_allox_alpha_parsing(self):
         assert get_policy("allox_alpha=0.7")._alpha == pytest.approx(0.7)
         assert get_policy("allox")._alpha == pytest.approx(0.2)
+
+
+class TestIsolatedPlus:
+    def test_no_scale_factor_division(self):
+        """isolated_plus gives a 4-GPU job the same TIME share as a
+        1-GPU job (isolated divides it by the scale factor)."""
+        from shockwave_amd.policies import (
+            IsolatedPolicy, IsolatedPlusPolicy,
+        )
+
+        tputs = mk_tputs([1.0, 1.0])
+        sf = {JobIdPair(0): 4, JobIdPair(1): 1}
+        plus = IsolatedPlusPolicy().get_allocation(tputs, sf, {WT: 2})
+        base = IsolatedPolicy().get_allocation(tputs, sf, {WT: 2})
+        assert plus[JobIdPair(0)][WT] == pytest.approx(
+            plus[JobIdPair(1)][WT]
+        )
+        assert base[JobIdPair(0)][WT] < base[JobIdPair(1)][WT]
