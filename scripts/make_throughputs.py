@@ -18,14 +18,12 @@ the product of their GPU utilizations.
 """
 
 import json
-import math
 import os
 import sys
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 from shockwave_amd.core import datasets
-from shockwave_amd.core.job_table import build_job_table
 from shockwave_amd.core.throughputs import format_job_type_key
 
 # Estimated MI355X single-GPU steps/s per (model, batch size).

@@ -14,8 +14,6 @@ import time
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 
-import numpy as np
-
 from shockwave_amd.core.job import JobIdPair
 from shockwave_amd.policies import get_policy
 from shockwave_amd.solver import PlannerJob, solve_eg_milp

@@ -9,7 +9,6 @@ import argparse
 import glob
 import os
 import pickle
-import sys
 
 import matplotlib
 
