@@ -112,7 +112,9 @@ def main(out_path="traces/mi355x_throughputs.json"):
                 e[format_job_type_key((jt2, 1))] = [mine, theirs]
         raw[WORKER_TYPE][format_job_type_key((jt, sf))] = e
 
-    os.makedirs(os.path.dirname(out_path), exist_ok=True)
+    out_dir = os.path.dirname(out_path)
+    if out_dir:
+        os.makedirs(out_dir, exist_ok=True)
     with open(out_path, "w") as f:
         json.dump(raw, f, indent=1)
     print(f"wrote {len(raw[WORKER_TYPE])} job-type entries -> {out_path}")
