@@ -183,3 +183,58 @@ class TestRelaxedPriorities:
         assert all(np.isfinite(pri))
         sched = p._greedy_schedule([0])
         assert sched.shape == (1, 3)
+
+
+class TestEGMilpProperties:
+    """Structural fairness/efficiency properties of the Nash-welfare MILP
+    (reference shockwave.py dynamic_eisenberg_gale_scheduling :504-712)."""
+
+    def test_symmetric_jobs_split_rounds_evenly(self):
+        jobs = [mk_planner_job(i, remaining=2000.0) for i in range(3)]
+        sol = solve_eg_milp(jobs, ngpus=1, round_index=0, future_nrounds=6,
+                            round_duration=120, logapx_bases=BASES,
+                            logapx_origin=ORIGIN, k=1e-3, rhomax=1.0,
+                            enable_ftf=False, timeout=30)
+        counts = sol.schedule.sum(axis=1)
+        assert counts.sum() <= 6
+        # identical jobs must not differ by more than one round
+        assert counts.max() - counts.min() <= 1
+
+    def test_higher_priority_gets_no_fewer_rounds(self):
+        jobs = [mk_planner_job(0, prio=5.0, remaining=2000.0),
+                mk_planner_job(1, prio=1.0, remaining=2000.0)]
+        sol = solve_eg_milp(jobs, ngpus=1, round_index=0, future_nrounds=4,
+                            round_duration=120, logapx_bases=BASES,
+                            logapx_origin=ORIGIN, k=1e-3, rhomax=1.0,
+                            enable_ftf=False, timeout=30)
+        counts = sol.schedule.sum(axis=1)
+        assert counts[0] >= counts[1]
+
+    def test_nearly_done_job_not_overscheduled(self):
+        """A job that finishes inside one round cannot usefully consume
+        more rounds than it needs; the slack goes to the long job."""
+        jobs = [mk_planner_job(0, epochs=10, progress=9, dur=10.0,
+                               remaining=10.0),
+                mk_planner_job(1, remaining=5000.0)]
+        sol = solve_eg_milp(jobs, ngpus=1, round_index=0, future_nrounds=6,
+                            round_duration=120, logapx_bases=BASES,
+                            logapx_origin=ORIGIN, k=1e-3, rhomax=1.0,
+                            enable_ftf=False, timeout=30)
+        counts = sol.schedule.sum(axis=1)
+        assert counts[1] >= 4  # long job gets the bulk of the horizon
+
+    def test_planned_progress_bounded(self):
+        """planned_progress is epochs gained over the horizon: at most
+        scheduled-time/epoch-duration and at most the remaining epochs."""
+        jobs = [mk_planner_job(i, epochs=10, progress=0, dur=100.0,
+                               remaining=900.0) for i in range(2)]
+        sol = solve_eg_milp(jobs, ngpus=2, round_index=0, future_nrounds=5,
+                            round_duration=120, logapx_bases=BASES,
+                            logapx_origin=ORIGIN, k=1e-3, rhomax=1.0,
+                            enable_ftf=False, timeout=30)
+        counts = sol.schedule.sum(axis=1)
+        for i in range(2):
+            assert sol.planned_progress[i] >= -1e-6
+            assert sol.planned_progress[i] <= min(
+                counts[i] * 120.0 / 100.0, 10.0
+            ) + 1e-6
