@@ -314,6 +314,28 @@ class TestJobsToComplete:
 
 
 class TestSweepHarness:
+    def test_run_sweep_static(self, tmp_path):
+        """Static mode: fixed batch at t=0, makespan recorded."""
+        import json as _json
+        import subprocess
+
+        out = subprocess.run(
+            [sys.executable,
+             os.path.join(os.path.dirname(__file__), "..", "scripts",
+                          "sweeps", "run_sweep.py"),
+             "--mode", "static", "--policies", "fifo",
+             "--num_jobs", "6", "-c", "4", "--seeds", "0", "-p", "1",
+             "--max_duration", "2000", "-l", str(tmp_path)],
+            capture_output=True, text=True, timeout=240,
+        )
+        assert out.returncode == 0, out.stderr[-2000:]
+        lines = [
+            _json.loads(line)
+            for line in open(tmp_path / "sweep_results.jsonl")
+        ]
+        assert len(lines) == 1 and lines[0]["status"] == "ok"
+        assert lines[0]["makespan_s"] > 0
+
     def test_run_sweep_continuous(self, tmp_path):
         """scripts/sweeps/run_sweep.py (reference run_sweep_continuous):
         grid runs in a process pool, results land in jsonl."""
