@@ -33,6 +33,7 @@ def run_simulation(
     results_dir=None,
     log_level="WARNING",
     preemption_overhead_s=20.0,
+    ideal=False,
 ):
     import logging
 
@@ -83,7 +84,8 @@ def run_simulation(
     )
 
     start = time.time()
-    makespan = sched.simulate({worker_type: num_gpus}, arrival_times, jobs)
+    makespan = sched.simulate({worker_type: num_gpus}, arrival_times, jobs,
+                              ideal=ideal)
     wall = time.time() - start
 
     avg_jct, geo_jct, har_jct, jct_list = sched.get_average_jct()
@@ -132,6 +134,9 @@ def run_simulation(
 def main():
     p = argparse.ArgumentParser(description=__doc__)
     p.add_argument("-t", "--trace_file", default=None)
+    p.add_argument("--ideal", action="store_true",
+                   help="round-free fractional-allocation upper bound "
+                        "(reference simulate ideal=True)")
     p.add_argument("--generate_jobs", type=int, default=None,
                    help="generate N jobs on the fly instead of a trace "
                         "(reference simulate_scheduler_with_generated_jobs)")
@@ -181,6 +186,7 @@ def main():
         results_dir=args.results_dir,
         log_level=args.log_level,
         preemption_overhead_s=args.preemption_overhead,
+        ideal=args.ideal,
     )
     print(
         json.dumps(

@@ -1205,10 +1205,108 @@ class RoundScheduler:
     # Simulation loop (reference simulate() :1728-2268)
     # ------------------------------------------------------------------
 
+    def _simulate_ideal(self, cluster_spec, arrival_times, jobs,
+                        jobs_to_complete=None):
+        """Round-free upper bound: between events, every job runs at
+        exactly its FRACTIONAL allocation (continuous time-sharing, no
+        rounds, no preemption cost) — the reference's ``ideal=True`` sim
+        branch (scheduler.py:2124-2180).  Adaptation twins are skipped:
+        this mode is an allocation-quality upper bound, not a mechanism
+        simulation."""
+        queued = list(zip(arrival_times, jobs))
+        for worker_type in sorted(cluster_spec):
+            for _ in range(cluster_spec[worker_type]):
+                self.register_worker(worker_type, num_gpus=1)
+        if queued:
+            self._current_timestamp = queued[0][0]
+        wt = self._worker_type
+        start_time = self._current_timestamp
+        EPS = 1e-6
+        while self._jobs or queued:
+            if jobs_to_complete is not None and self.is_done(jobs_to_complete):
+                break
+            while queued and queued[0][0] <= self._current_timestamp + EPS:
+                _, job = queued.pop(0)
+                self.add_job(job, timestamp=self._current_timestamp)
+            if not self._jobs:
+                if queued:
+                    self._current_timestamp = queued[0][0]
+                    continue
+                break
+            self._need_to_update_allocation = True
+            self._last_reset_time = (
+                -self._minimum_time_between_allocation_resets
+            )
+            self._allocation = self._compute_allocation()
+            # aggregate per-single rates (steps/s) over all allocation rows
+            rates = {}
+            time_fracs = {}
+            for jid, per_wt in (self._allocation or {}).items():
+                frac = per_wt.get(wt, 0.0)
+                if frac <= 0 or jid not in self._throughputs:
+                    continue
+                tput = self._throughputs[jid].get(wt)
+                if tput is None:
+                    continue
+                time_fracs[jid] = frac
+                if jid.is_pair():
+                    for k, s in enumerate(jid.singletons()):
+                        if s in self._jobs:
+                            rates[s] = rates.get(s, 0.0) + frac * tput[k]
+                else:
+                    rates[jid] = rates.get(jid, 0.0) + frac * tput
+            next_arrival = queued[0][0] if queued else None
+            dts = []
+            if next_arrival is not None:
+                dts.append(next_arrival - self._current_timestamp)
+            for jid, rate in rates.items():
+                if rate > 0:
+                    dts.append(self._get_remaining_steps(jid) / rate)
+            dts = [d for d in dts if d > 0]
+            if not dts:
+                if next_arrival is not None:
+                    self._current_timestamp = next_arrival
+                    continue
+                for jid in list(self._jobs):  # starved forever: fail out
+                    logger.error("ideal sim: job %s unschedulable", jid)
+                    self._per_job_latest_timestamps[jid] = (
+                        self._current_timestamp
+                    )
+                    self._num_failures_per_job[jid] = MAX_FAILED_ATTEMPTS
+                    self._remove_job(jid)
+                break
+            dt = max(EPS, min(dts))
+            self._current_timestamp += dt
+            for jid, frac in time_fracs.items():
+                if jid in self._job_time_so_far:
+                    self._job_time_so_far[jid][wt] = (
+                        self._job_time_so_far[jid].get(wt, 0.0) + frac * dt
+                    )
+            for jid, rate in rates.items():
+                self._total_steps_run[jid] += rate * dt
+                self._steps_run_so_far[jid][wt] = (
+                    self._steps_run_so_far[jid].get(wt, 0) + rate * dt
+                )
+                self._per_job_latest_timestamps[jid] = (
+                    self._current_timestamp
+                )
+                if self._get_remaining_steps(jid) <= 0.5:
+                    logger.info("[Job succeeded] (ideal) %s", jid)
+                    self._remove_job(jid)
+        makespan = self._current_timestamp - start_time
+        logger.info("ideal simulation complete: makespan %.1f s", makespan)
+        return makespan
+
     def simulate(self, cluster_spec, arrival_times, jobs,
                  num_gpus_per_server=None, debug=False,
                  checkpoint_threshold=None, checkpoint_file=None,
-                 jobs_to_complete=None, _resume_state=None):
+                 jobs_to_complete=None, ideal=False, _resume_state=None):
+        if ideal:
+            assert _resume_state is None, "ideal mode has no checkpoints"
+            return self._simulate_ideal(
+                cluster_spec, arrival_times, jobs,
+                jobs_to_complete=jobs_to_complete,
+            )
         if _resume_state is None:
             queued_jobs = list(zip(arrival_times, jobs))
             remaining_jobs = len(jobs)

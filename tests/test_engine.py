@@ -482,3 +482,15 @@ class TestCostAndSLOMetrics:
             slo=36000.0,
         )
         assert len(sched.get_job_completion_times()) == 8
+
+
+class TestIdealMode:
+    def test_ideal_upper_bounds_round_sim(self, throughputs):
+        """ideal=True (continuous fractional allocation, reference
+        scheduler.py:2124-2180) completes all jobs with makespan <= the
+        round-based mechanism's."""
+        r_round = sim("max_min_fairness")
+        r_ideal = run_simulation(TRACE_8, ORACLE, "max_min_fairness",
+                                 num_gpus=2, ideal=True)
+        assert len(r_ideal["jct_list"]) == 8
+        assert r_ideal["makespan_s"] <= r_round["makespan_s"] * 1.001
