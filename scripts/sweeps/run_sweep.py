@@ -102,6 +102,7 @@ def run_experiment(spec):
             list(arrivals),
             copy.deepcopy(jobs),
             jobs_to_complete=window,
+            ideal=spec.get("ideal", False),
         )
         window_ids = sorted(window) if window else None
         avg_jct, geo_jct, _, _ = sched.get_average_jct(window_ids)
@@ -164,6 +165,7 @@ def build_grid(args):
                         "min_duration_s": args.min_duration,
                         "max_duration_s": args.max_duration,
                         "throughputs_file": args.throughputs_file,
+                        "ideal": args.ideal,
                         "timeout": args.timeout,
                     }
                 )
@@ -203,6 +205,8 @@ def main():
     p.add_argument("--max_duration", type=float, default=8000.0)
     p.add_argument("-p", "--processes", type=int,
                    default=max(1, (os.cpu_count() or 2) // 2))
+    p.add_argument("--ideal", action="store_true",
+                   help="round-free fractional-allocation upper bound")
     p.add_argument("--timeout", type=float, default=1200.0,
                    help="per-experiment timeout (s)")
     p.add_argument("-l", "--log_dir", default="results/sweep")
