@@ -527,6 +527,9 @@ class RoundScheduler:
             allocation = self._policy.get_allocation(
                 throughputs, scale_factors, state["priority_weights"], cluster_spec
             )
+            if isinstance(allocation, tuple):
+                # strategy-proof perf returns (allocation, discounts)
+                allocation = allocation[0]
         elif name.startswith("MinTotalDuration"):
             allocation = self._policy.get_allocation(
                 throughputs, scale_factors, state["num_steps_remaining"], cluster_spec

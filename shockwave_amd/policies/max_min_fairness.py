@@ -451,12 +451,13 @@ class MaxMinFairnessStrategyProofPolicyWithPerf(Policy):
             for jid in job_ids:
                 minus = dict(unflattened_throughputs)
                 del minus[jid]
-                rates_minus_job.append(
-                    self.get_allocation(
-                        minus, scale_factors, unflattened_priority_weights,
-                        cluster_spec, recurse_deeper=False,
-                    )
+                # with a single job the leave-one-out economy is empty:
+                # no externality, no discount
+                rates = self.get_allocation(
+                    minus, scale_factors, unflattened_priority_weights,
+                    cluster_spec, recurse_deeper=False,
                 )
+                rates_minus_job.append(rates if rates is not None else {})
 
         priority = np.array(
             [1.0 / unflattened_priority_weights[jid] for jid in job_ids]

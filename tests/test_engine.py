@@ -494,3 +494,12 @@ class TestIdealMode:
                                  num_gpus=2, ideal=True)
         assert len(r_ideal["jct_list"]) == 8
         assert r_ideal["makespan_s"] <= r_round["makespan_s"] * 1.001
+
+
+class TestStrategyProofPerfSim:
+    def test_sim_completes(self, throughputs):
+        """Engine unwraps the (allocation, discounts) tuple the
+        strategy-proof perf policy returns; single-job leave-one-out
+        economies are empty, not None (regression from policy soak)."""
+        r = sim("max_min_fairness_strategy_proof_perf")
+        assert len(r["jct_list"]) == 8
