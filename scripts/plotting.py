@@ -75,13 +75,52 @@ def plot_per_round_schedule(result, out_path, max_rounds=400):
     plt.close()
 
 
+def plot_load_sweep(jsonl_paths, out_path):
+    """Steady-state avg JCT vs input load from run_sweep jsonl rows."""
+    import collections
+    import json
+
+    rows = []
+    for p in jsonl_paths:
+        with open(p) as f:
+            rows += [json.loads(line) for line in f]
+    agg = collections.defaultdict(list)
+    for r in rows:
+        if r.get("status") == "ok":
+            agg[(r["policy"], r["jobs_per_hr"])].append(
+                r["avg_jct_s"] / 3600.0
+            )
+    pols = sorted({p for p, _ in agg})
+    loads = sorted({l for _, l in agg})
+    fig, ax = plt.subplots(figsize=(7, 4.5))
+    for pol in pols:
+        ys = [np.mean(agg[(pol, l)]) for l in loads if (pol, l) in agg]
+        xs = [l for l in loads if (pol, l) in agg]
+        kw = ({"linewidth": 2.5, "marker": "o"} if pol == "shockwave"
+              else {"linewidth": 1.2, "marker": ".", "alpha": 0.8})
+        ax.plot(xs, ys, label=pol, **kw)
+    ax.set_xlabel("input load (jobs/hr)")
+    ax.set_ylabel("steady-state avg JCT (h)")
+    ax.grid(alpha=0.3)
+    ax.legend(fontsize=8)
+    fig.tight_layout()
+    fig.savefig(out_path, dpi=130)
+    plt.close(fig)
+
+
 def main():
     p = argparse.ArgumentParser(description=__doc__)
     p.add_argument("--results_dir", required=True)
     p.add_argument("--out_dir", default=None)
+    p.add_argument("--load_sweep_jsonl", nargs="*", default=None,
+                   help="run_sweep result files -> load_sweep.png")
     args = p.parse_args()
     out_dir = args.out_dir or args.results_dir
     os.makedirs(out_dir, exist_ok=True)
+
+    if args.load_sweep_jsonl:
+        plot_load_sweep(args.load_sweep_jsonl,
+                        os.path.join(out_dir, "load_sweep.png"))
 
     results = load_results(args.results_dir)
     if not results:
