@@ -99,3 +99,105 @@ class TestUpdateBsAllTemplates:
         job.update_bs(old_bs * 2)
         assert job.batch_size == old_bs * 2
         assert str(old_bs * 2) in job.command
+
+
+@needs_reference
+class TestReferenceParity:
+    """Run OUR simulator on the reference's own trace + V100 oracle and
+    compare against the result pickles the reference repo ships
+    (reproduce/pickles/tacc_32gpus) — profiles/REFERENCE_PARITY.md."""
+
+    TRACE = os.path.join(
+        REFERENCE, "traces", "reproduce",
+        "120_0.2_5_100_40_25_0,0.5,0.5_0.6,0.3,0.09,0.01"
+        "_multigpu_dynamic.trace",
+    )
+    PICKLES = os.path.join(REFERENCE, "reproduce", "pickles", "tacc_32gpus")
+
+    @staticmethod
+    def _load_reference_pickle(policy):
+        import pickle
+
+        class _RefJobIdPair:
+            def __init__(self, *a, **k):
+                pass
+
+            def __hash__(self):
+                return hash((self.__dict__.get("_job0"),
+                             self.__dict__.get("_job1")))
+
+            def __eq__(self, o):
+                return (isinstance(o, _RefJobIdPair)
+                        and self.__dict__ == o.__dict__)
+
+        class RefUnpickler(pickle.Unpickler):
+            def find_class(self, module, name):
+                if module == "job_id_pair" and name == "JobIdPair":
+                    return _RefJobIdPair
+                return super().find_class(module, name)
+
+        path = os.path.join(
+            TestReferenceParity.PICKLES,
+            f"{policy}_120_0.2_5_100_40_25_0,0.5,0.5_0.6,0.3,0.09,0.01"
+            "_multigpu_dynamic_simulation.pickle",
+        )
+        with open(path, "rb") as f:
+            return RefUnpickler(f).load()
+
+    def _run_ours(self, policy, shockwave_config=None):
+        import copy
+
+        from shockwave_amd.core.throughputs import read_throughputs
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        tputs = read_throughputs(
+            os.path.join(REFERENCE, "tacc_throughputs.json")
+        )
+        jobs, arrivals = trace_mod.parse_trace(self.TRACE)
+        profiles = [trace_mod.build_job_profile(j, tputs) for j in jobs]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+        sched = RoundScheduler(
+            get_policy(policy), simulate=True, throughputs=tputs,
+            time_per_iteration=120, profiles=profiles,
+            shockwave_config=shockwave_config, worker_type="v100",
+            seed=0, preemption_overhead_s=20.0,
+        )
+        makespan = sched.simulate({"v100": 32}, list(arrivals),
+                                  copy.deepcopy(jobs))
+        return sched, makespan
+
+    def test_max_min_fairness_matches_reference_exactly(self):
+        """Deterministic greedy mechanism on identical inputs: makespan
+        within 0.5%, avg JCT and worst rho within 2%."""
+        ref = self._load_reference_pickle("max_min_fairness")
+        sched, makespan = self._run_ours("max_min_fairness")
+        assert len(sched.get_job_completion_times()) == 120
+        assert abs(makespan - ref["makespan"]) / ref["makespan"] < 0.005
+        assert (abs(sched.get_average_jct()[0] - float(ref["avg_jct"]))
+                / float(ref["avg_jct"]) < 0.02)
+        ftf, _ = sched.get_finish_time_fairness()
+        ref_rho = max(ref["finish_time_fairness_list"])
+        assert abs(max(ftf) - ref_rho) / ref_rho < 0.02
+
+    def test_allox_matches_reference(self):
+        ref = self._load_reference_pickle("allox")
+        sched, makespan = self._run_ours("allox")
+        assert abs(makespan - ref["makespan"]) / ref["makespan"] < 0.01
+
+    def test_shockwave_matches_reference(self):
+        """Independent EG MILP (scipy/HiGHS vs Gurobi): makespan and avg
+        JCT within 5% of the reference's committed shockwave result."""
+        import json
+
+        cfg = json.load(open(os.path.join(
+            REFERENCE, "configurations", "tacc_32gpus.json")))
+        cfg["time_per_iteration"] = 120
+        cfg["num_gpus"] = 32
+        ref = self._load_reference_pickle("shockwave")
+        sched, makespan = self._run_ours("shockwave", shockwave_config=cfg)
+        assert len(sched.get_job_completion_times()) == 120
+        assert abs(makespan - ref["makespan"]) / ref["makespan"] < 0.05
+        assert (abs(sched.get_average_jct()[0] - float(ref["avg_jct"]))
+                / float(ref["avg_jct"]) < 0.05)
