@@ -495,6 +495,31 @@ class TestIdealMode:
         assert len(r_ideal["jct_list"]) == 8
         assert r_ideal["makespan_s"] <= r_round["makespan_s"] * 1.001
 
+    def test_ideal_with_measurement_window(self, throughputs):
+        """ideal mode honors jobs_to_complete (the sweep's windowed
+        steady-state path)."""
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.core.job import JobIdPair
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+        sched = RoundScheduler(
+            get_policy("max_min_fairness"), simulate=True,
+            throughputs=throughputs, time_per_iteration=120,
+            profiles=profiles, worker_type="mi355x",
+        )
+        window = {JobIdPair(0), JobIdPair(1)}
+        sched.simulate({"mi355x": 2}, arrivals, jobs, ideal=True,
+                       jobs_to_complete=window)
+        done = set(sched.get_job_completion_times())
+        assert window <= done
+        assert len(done) < 8
+
 
 class TestStrategyProofPerfSim:
     def test_sim_completes(self, throughputs):
