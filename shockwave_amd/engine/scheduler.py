@@ -1351,8 +1351,10 @@ class RoundScheduler:
                 current_round_end_time = max_timestamp
             if max_timestamp > 0:
                 self._current_timestamp = max_timestamp
-            else:
+            elif next_job_arrival_time is not None:
                 self._current_timestamp = next_job_arrival_time
+            # else: resumed at a round boundary with every job admitted and
+            # nothing in flight — keep the restored clock
 
             # drain completed micro-tasks
             while running_jobs:

@@ -528,3 +528,36 @@ class TestStrategyProofPerfSim:
         economies are empty, not None (regression from policy soak)."""
         r = sim("max_min_fairness_strategy_proof_perf")
         assert len(r["jct_list"]) == 8
+
+
+class TestResumeAfterAllArrivals:
+    def test_resume_with_empty_arrival_queue(self, tmp_path, throughputs):
+        """A checkpoint taken AFTER every job has been admitted resumes
+        with an empty arrival queue and nothing in flight: the clock must
+        keep its restored value instead of becoming None (regression:
+        TypeError in _update_priorities on resume)."""
+        import copy
+
+        from shockwave_amd.core import trace as trace_mod
+        from shockwave_amd.engine import RoundScheduler
+        from shockwave_amd.policies import get_policy
+
+        jobs, arrivals = trace_mod.parse_trace(TRACE_8)
+        profiles = [trace_mod.build_job_profile(j, throughputs)
+                    for j in jobs]
+        for j, pr in zip(jobs, profiles):
+            j.duration = sum(pr["duration_every_epoch"])
+        ckpt = str(tmp_path / "late.ckpt")
+        sched = RoundScheduler(
+            get_policy("fifo"), simulate=True, throughputs=throughputs,
+            time_per_iteration=120, profiles=copy.deepcopy(profiles),
+            worker_type="mi355x",
+        )
+        # threshold = total job count: the checkpoint lands once the last
+        # job has arrived, i.e. with an empty queue
+        sched.simulate({"mi355x": 2}, list(arrivals),
+                       copy.deepcopy(jobs), checkpoint_threshold=8,
+                       checkpoint_file=ckpt)
+        resumed, state = RoundScheduler.resume_simulation(ckpt)
+        resumed.simulate({"mi355x": 2}, None, None, _resume_state=state)
+        assert len(resumed.get_job_completion_times()) == 8
