@@ -44,9 +44,17 @@ class WorkloadSpec:
     copy_batch: Callable = None          # (static_batch, batch) -> None
 
 
+class _Preempted(SystemExit):
+    """Raised by the SIGTERM handler: the dispatcher's watchdog kill sends
+    SIGTERM 10 s before SIGKILL (dispatcher.kill_job) — enough to
+    checkpoint and flush progress so the round's steps are not lost.
+    (The reference kills by SIGKILL and loses the round's work.)"""
+
+
 def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
         max_steps_override: Optional[int] = None):
     import os
+    import signal
 
     mode = args.mode or mode or os.environ.get("SWQ_MODE", "static")
     device = common.init_device_and_distributed(args)
@@ -193,25 +201,45 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
 
         graphed = try_graph_step(lambda: graph_body(), [])
 
+    def _on_sigterm(signum, frame):
+        raise _Preempted(0)
+
+    prev_sigterm = None
+    try:
+        prev_sigterm = signal.signal(signal.SIGTERM, _on_sigterm)
+    except ValueError:
+        pass  # not the main thread (in-process tests)
+
     done = False
-    while not done and state["cumulative_steps"] < target_steps:
-        hit_target = False
-        for batch in trainloader:
-            if graphed is not None:
-                spec.copy_batch(static_batch, batch)
-                graphed.replay()
-            else:
-                eager_step(batch)
-            state["cumulative_steps"] += 1
-            reporter.step()
-            if state["cumulative_steps"] >= target_steps:
-                hit_target = True
+    try:
+        while not done and state["cumulative_steps"] < target_steps:
+            hit_target = False
+            for batch in trainloader:
+                if graphed is not None:
+                    spec.copy_batch(static_batch, batch)
+                    graphed.replay()
+                else:
+                    eager_step(batch)
+                state["cumulative_steps"] += 1
+                reporter.step()
+                if state["cumulative_steps"] >= target_steps:
+                    hit_target = True
+                    break
+            if hit_target or (lease_it is not None and lease_it.done):
                 break
-        if hit_target or (lease_it is not None and lease_it.done):
-            break
-        if maybe_request_rescale(state["epoch"]):
-            done = True
-        state["epoch"] += 1
+            if maybe_request_rescale(state["epoch"]):
+                done = True
+            state["epoch"] += 1
+    except _Preempted:
+        # watchdog kill: fall through to checkpoint + progress flush
+        # below, then exit cleanly inside the dispatcher's SIGTERM window
+        pass
+    finally:
+        if prev_sigterm is not None:
+            try:
+                signal.signal(signal.SIGTERM, prev_sigterm)
+            except ValueError:
+                pass
 
     if lease_it is not None:
         save_checkpoint()

@@ -264,3 +264,52 @@ class TestRecommendationUnit:
         n_late = ndcg_at_k(worst_in_k, targets, k=6)[0].item()
         assert n_best == pytest.approx(1.0)
         assert 0 < n_late < n_best
+
+
+class TestSigtermGracefulPreemption:
+    def test_sigterm_checkpoints_and_flushes_progress(self, tmp_path):
+        """The dispatcher's watchdog kill sends SIGTERM 10 s before
+        SIGKILL; the training loop must catch it, checkpoint, flush
+        PROGRESS lines, and exit 0 — so the round's steps are scraped
+        instead of lost (the reference SIGKILLs and drops the round)."""
+        import signal
+        import time
+
+        env = dict(os.environ)
+        env.update({
+            "SWQ_DATASET_LEN": "50000",
+            "SWQ_GRAPHS": "0",
+            "GAVEL_JOB_ID": "-1",
+            "GAVEL_WORKER_ID": "0",
+            "GAVEL_ROUND_ID": "0",
+        })
+        ckpt = str(tmp_path / "job")
+        cmd = [
+            sys.executable, "main.py", "--batch_size", "16",
+            "--num_steps", "100000", "--checkpoint_dir", ckpt,
+            "--enable_gavel_iterator",
+        ]
+        cwd = os.path.join(REPO, "workloads", "pytorch",
+                           "image_classification", "cifar10")
+        proc = subprocess.Popen(cmd, cwd=cwd, env=env,
+                                stdout=subprocess.PIPE,
+                                stderr=subprocess.STDOUT)
+        # wait until it has made some steps (first log flush proves the
+        # iterator is up), then preempt
+        time.sleep(25)
+        proc.send_signal(signal.SIGTERM)
+        out, _ = proc.communicate(timeout=60)
+        assert proc.returncode == 0, out.decode()[-2000:]
+        # checkpoint written
+        import glob
+
+        assert glob.glob(os.path.join(ckpt, "*.chkpt")) or glob.glob(
+            os.path.join(ckpt, "*.pt")
+        ) or glob.glob(os.path.join(ckpt, "model*")), os.listdir(ckpt)
+        # progress lines flushed with nonzero steps
+        log = os.path.join(ckpt, ".gavel", "round=0", "worker=0.log")
+        steps = 0
+        for line in open(log):
+            if "[PROGRESS] [STEPS]" in line:
+                steps = int(float(line.rsplit("]", 1)[1].strip()))
+        assert steps > 0, open(log).read()[-1500:]
