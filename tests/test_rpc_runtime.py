@@ -373,3 +373,43 @@ class TestPackedDispatch:
                                 round_id=0)
         got = {jid: steps for jid, steps, _, _ in notified["jobs"]}
         assert got == {1: 11, 2: 22}
+
+
+class TestGpuDiscoveryAndLogging:
+    def test_get_num_gpus_cpu_box(self):
+        """On a GPU-less box discovery degrades to 0 (or a real count if
+        smi tools are present) without raising."""
+        from shockwave_amd.runtime.gpu import get_num_gpus
+
+        n = get_num_gpus()
+        assert isinstance(n, int) and n >= 0
+
+    def test_gpu_processes_best_effort(self):
+        from shockwave_amd.runtime.gpu import get_gpu_processes
+
+        procs = get_gpu_processes()
+        assert isinstance(procs, dict)
+
+    def test_scheduler_adapter_prefixes_clock(self):
+        """SchedulerAdapter prefixes records with the scheduler clock
+        (reference custom_logging.py:5-13)."""
+        import logging as _logging
+
+        from shockwave_amd.utils.logging import SchedulerAdapter
+
+        class FakeSched:
+            def get_current_timestamp(self):
+                return 1234.5
+
+        records = []
+
+        class Capture(_logging.Handler):
+            def emit(self, record):
+                records.append(record.getMessage())
+
+        lg = _logging.getLogger("swq.adapter.test")
+        lg.setLevel(_logging.INFO)
+        lg.addHandler(Capture())
+        SchedulerAdapter(lg, FakeSched()).info("round %d done", 7)
+        assert records and records[0].startswith("[1234.50] ")
+        assert "round 7 done" in records[0]
