@@ -23,11 +23,14 @@ collectives) instead.
 from __future__ import annotations
 
 import io
+import logging
 import os
 import shutil
 from typing import Optional
 
 import torch
+
+logger = logging.getLogger("shockwave_amd.ckpt_stream")
 
 
 class CheckpointStore:
@@ -98,7 +101,16 @@ class CheckpointStore:
 
     def load(self) -> Optional[dict]:
         for path in self._read_order():
-            return torch.load(path, map_location="cpu", weights_only=False)
+            try:
+                return torch.load(path, map_location="cpu",
+                                  weights_only=False)
+            except Exception:
+                # corrupt tier (e.g. node died mid-write before the
+                # atomic rename landed everywhere): fall through to the
+                # next tier; a fresh start beats a crashed job
+                logger.exception("corrupt checkpoint at %s; trying next "
+                                 "tier", path)
+                continue
         return None
 
     def read_bytes(self) -> Optional[bytes]:
