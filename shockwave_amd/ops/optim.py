@@ -142,7 +142,20 @@ class FusedAdam(torch.optim.Optimizer):
         return d
 
     def load_state_dict(self, state_dict):
-        self._step_count = state_dict.pop("swq_step_count", 0)
+        # read without mutating the caller's dict; a checkpoint written by
+        # stock torch.optim.Adam carries per-param 'step' entries instead —
+        # derive the count from those so bias correction resumes correctly
+        if "swq_step_count" in state_dict:
+            self._step_count = state_dict["swq_step_count"]
+            state_dict = {k: v for k, v in state_dict.items()
+                          if k != "swq_step_count"}
+        else:
+            steps = [
+                int(s["step"].item() if torch.is_tensor(s.get("step"))
+                    else s.get("step", 0))
+                for s in state_dict.get("state", {}).values()
+            ]
+            self._step_count = max(steps) if steps else 0
         super().load_state_dict(state_dict)
         self._cached_lists = None
 

@@ -115,8 +115,18 @@ class CheckpointStore:
 
     def read_bytes(self) -> Optional[bytes]:
         for path in self._read_order():
-            with open(path, "rb") as f:
-                return f.read()
+            try:
+                with open(path, "rb") as f:
+                    data = f.read()
+                # validate before serving over FetchCheckpoint: a torn
+                # /dev/shm file must not shadow an intact durable copy
+                torch.load(io.BytesIO(data), map_location="cpu",
+                           weights_only=False)
+                return data
+            except Exception:
+                logger.exception("corrupt checkpoint at %s; trying next "
+                                 "tier", path)
+                continue
         return None
 
     def write_bytes(self, data: bytes) -> None:

@@ -76,20 +76,26 @@ class BucketedDataParallel(torch.nn.Module):
                 p.register_post_accumulate_grad_hook(self._grad_ready_hook)
             )
         self._require_finish = False
+        self._accumulating = False
 
     def _make_bucket(self, params: List[torch.nn.Parameter]):
-        total = sum(p.numel() for p in params)
         device = params[0].device
         dtype = params[0].dtype
-        buffer = torch.zeros(total, dtype=dtype, device=device)
+        # the fused optimizer kernels (ops/csrc/fused_ops.hip) vectorize as
+        # float4 (16 B); every grad view must start 16-B aligned, so round
+        # each param's offset up to an element multiple of 16 bytes
+        align = max(1, 16 // params[0].element_size())
+        offsets = []
         offset = 0
         for p in params:
-            n = p.numel()
+            offsets.append(offset)
+            offset += -(-p.numel() // align) * align
+        buffer = torch.zeros(offset, dtype=dtype, device=device)
+        for p, off in zip(params, offsets):
             # match the param's own dense layout (NCHW or channels_last) so
             # autograd accumulates into the bucket with the param's strides
             # and the flat fused-optimizer kernels see aligned layouts
-            p.grad = buffer.as_strided(p.shape, p.stride(), offset)
-            offset += n
+            p.grad = buffer.as_strided(p.shape, p.stride(), off)
         bucket = Bucket(params, buffer)
         for p in params:
             self._param_bucket[p] = bucket
@@ -109,6 +115,7 @@ class BucketedDataParallel(torch.nn.Module):
             bucket.pending == 0
             and self.world_size > 1
             and self.sync_mode == "hook"
+            and not self._accumulating
         ):
             bucket.work = dist.all_reduce(
                 bucket.buffer, op=dist.ReduceOp.SUM, group=self.group,
@@ -116,12 +123,32 @@ class BucketedDataParallel(torch.nn.Module):
             )
 
     def forward(self, *args, **kwargs):
-        if torch.is_grad_enabled():
+        # arm the countdown lazily: a second grad-enabled forward before
+        # finish_gradient_sync (GAN-style multi-forward, grad accumulation)
+        # must NOT reset pending mid-flight, or a bucket could all-reduce a
+        # partial gradient
+        if torch.is_grad_enabled() and not self._require_finish:
             for b in self.buckets:
                 b.pending = len(b.params)
                 b.work = None
             self._require_finish = True
         return self.module(*args, **kwargs)
+
+    def no_sync(self):
+        """Context manager: skip hook-triggered all-reduce (gradient
+        accumulation).  The final (synchronizing) step runs outside it."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def _ctx():
+            prev = self._accumulating
+            self._accumulating = True
+            try:
+                yield
+            finally:
+                self._accumulating = prev
+
+        return _ctx()
 
     def finish_gradient_sync(self):
         """Call after backward, before optimizer.step()."""
@@ -144,12 +171,12 @@ class BucketedDataParallel(torch.nn.Module):
                     w.wait()
             else:
                 for b in self.buckets:
-                    if b.pending == 0 and b.work is not None:
+                    if b.work is not None:
                         b.work.wait()
-                    elif b.pending > 0 and any(
-                        p.grad is not None for p in b.params
-                    ):
-                        # partial bucket (unused params): reduce now
+                    else:
+                        # the hook never fired the collective: partial
+                        # bucket (unused params), or hooks ran under
+                        # no_sync / multi-backward accumulation
                         dist.all_reduce(
                             b.buffer, op=dist.ReduceOp.SUM, group=self.group
                         )

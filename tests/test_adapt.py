@@ -129,3 +129,42 @@ class TestGNS:
         for p in model.parameters():
             p.grad.fill_(99.0)
         assert torch.equal(est._grads[0], first)
+
+
+class TestFusedAdamStateDict:
+    def test_load_from_stock_adam_preserves_step(self):
+        """resuming from a stock torch.optim.Adam checkpoint
+        must derive the bias-correction step from per-param 'step' state,
+        and must not mutate the caller's dict."""
+        import copy
+
+        import torch
+
+        from shockwave_amd.ops.optim import FusedAdam
+
+        model = torch.nn.Linear(4, 4)
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+        for _ in range(5):
+            model(torch.randn(2, 4)).sum().backward()
+            opt.step()
+            opt.zero_grad()
+        sd = opt.state_dict()
+        sd_copy = copy.deepcopy(sd)
+        fused = FusedAdam(model.parameters(), lr=1e-3)
+        fused.load_state_dict(sd)
+        assert fused._step_count == 5
+        assert sd.keys() == sd_copy.keys()  # caller's dict not mutated
+
+    def test_own_state_dict_roundtrip_no_mutation(self):
+        import torch
+
+        from shockwave_amd.ops.optim import FusedAdam
+
+        model = torch.nn.Linear(4, 4)
+        fused = FusedAdam(model.parameters(), lr=1e-3)
+        fused._step_count = 7
+        sd = fused.state_dict()
+        fused2 = FusedAdam(model.parameters(), lr=1e-3)
+        fused2.load_state_dict(sd)
+        assert fused2._step_count == 7
+        assert "swq_step_count" in sd  # caller's dict untouched

@@ -117,3 +117,21 @@ class TestCheckpointStoreEdgeCases:
         except Exception as e:
             pytest.fail(f"corrupt checkpoint must not raise: {e!r}")
         assert out is None
+
+    def test_read_bytes_skips_torn_shm_tier(self, tmp_path):
+        """a torn /dev/shm copy must not be served over
+        FetchCheckpoint when the durable copy is intact."""
+        import io
+
+        store = CheckpointStore(str(tmp_path / "job_id=12"),
+                                shm_root=str(tmp_path / "shm"))
+        store.save({"v": torch.tensor([4.0])})
+        # tear the shm copy (newer mtime than durable so it is preferred)
+        shm_path = store._paths()[0]
+        with open(shm_path, "r+b") as f:
+            f.truncate(16)
+        os.utime(shm_path)
+        data = store.read_bytes()
+        assert data is not None
+        out = torch.load(io.BytesIO(data), weights_only=False)
+        assert float(out["v"]) == 4.0

@@ -144,3 +144,59 @@ class TestManualSyncMode:
             torch.testing.assert_close(b, g)
         # calling again (replay-driven cadence) must not error
         ddp.finish_gradient_sync()
+
+
+class TestBucketAlignment:
+    def test_views_are_16B_aligned_with_odd_numels(self):
+        """params with numel%4!=0 (e.g. a bias of 10) must not
+        misalign the following grads — the fused kernels cast to float4*."""
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            store = dist.TCPStore("127.0.0.1", 0, 1, True)
+            dist.init_process_group("gloo", store=store, rank=0, world_size=1)
+        from shockwave_amd.parallel import BucketedDataParallel
+
+        model = torch.nn.Sequential(
+            torch.nn.Linear(7, 10),   # weight 70, bias 10 — both odd
+            torch.nn.Linear(10, 3),   # weight 30, bias 3
+        )
+        ddp = BucketedDataParallel(model, bucket_bytes=10**9)  # one bucket
+        for p in model.parameters():
+            assert p.grad.data_ptr() % 16 == 0, p.shape
+        # padding must not break correctness
+        x = torch.randn(4, 7)
+        ddp(x).sum().backward()
+        ddp.finish_gradient_sync()
+        ref = torch.nn.Sequential(
+            torch.nn.Linear(7, 10), torch.nn.Linear(10, 3)
+        )
+        ref.load_state_dict(model.state_dict())
+        ref(x).sum().backward()
+        for p, r in zip(model.parameters(), ref.parameters()):
+            torch.testing.assert_close(p.grad, r.grad)
+
+    def test_multi_forward_single_sync(self):
+        """GAN-style: two forwards, two backwards, then one sync — the
+        second forward must not reset the countdown mid-flight."""
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            store = dist.TCPStore("127.0.0.1", 0, 1, True)
+            dist.init_process_group("gloo", store=store, rank=0, world_size=1)
+        from shockwave_amd.parallel import BucketedDataParallel
+
+        torch.manual_seed(7)
+        model = torch.nn.Linear(6, 6)
+        ddp = BucketedDataParallel(model, bucket_bytes=16)
+        xa, xb = torch.randn(3, 6), torch.randn(3, 6)
+        la = ddp(xa).sum()
+        lb = ddp(xb).sum()    # second forward before any backward
+        la.backward()
+        lb.backward()
+        ddp.finish_gradient_sync()
+        ref = torch.nn.Linear(6, 6)
+        ref.load_state_dict(model.state_dict())
+        (ref(xa).sum() + ref(xb).sum()).backward()
+        for p, r in zip(model.parameters(), ref.parameters()):
+            torch.testing.assert_close(p.grad, r.grad)
