@@ -170,10 +170,11 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
             gns.on_step()
         optimizer.step()
 
-    # hipGraph the whole compute step for static-mode GPU jobs: the
-    # adaptation hooks of accordion (multi-tensor accumulate) would be
-    # capturable, but GNS's window snapshots allocate per step, so graphs
-    # are enabled for static jobs only
+    # hipGraph the whole compute step.  All three modes are capturable:
+    # accordion's multi-tensor accumulate is in-place, and GNS's window is
+    # a preallocated address-stable ring (adapt/gns.py) — include the
+    # adaptation hooks inside the captured body so one replay is the whole
+    # step.  (Epoch-boundary reads — norms, GNS scalar — happen outside.)
     graphed = None
     static_batch = None
     # capture costs seconds (warmup + MIOpen find on first process); only
@@ -183,7 +184,6 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
         expected_steps = min(expected_steps, lease_it._lease.max_steps)
     if (
         device.type == "cuda"
-        and mode == "static"
         and spec.make_static_batch is not None
         and expected_steps >= int(os.environ.get("SWQ_GRAPH_MIN_STEPS", "100"))
         and os.environ.get("SWQ_GRAPHS", "1") != "0"
@@ -197,9 +197,17 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
             loss = spec.step(model_train, static_batch, device, state)
             loss.backward()
             common.finish_sync(model_train)
+            if accordion is not None:
+                accordion.on_step()
+            if gns is not None:
+                gns.on_step()
             optimizer.step()
 
-        graphed = try_graph_step(lambda: graph_body(), [])
+        # GNS: the ring-fill and EMA-init branches must reach steady state
+        # before capture (gns.py docstring), so warm up window+1 steps
+        warmup = 3 if gns is None else max(3, gns.window + 1)
+        graphed = try_graph_step(lambda: graph_body(), [],
+                                 warmup_iters=warmup)
 
     def _on_sigterm(signum, frame):
         raise _Preempted(0)

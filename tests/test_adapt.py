@@ -118,17 +118,54 @@ class TestGNS:
         assert not est.should_double(7)   # before first decision window
         assert not est.should_double(17)  # not a decision epoch
 
-    def test_window_copies_not_views(self):
-        """Grad buffers are reused across steps; the window must snapshot."""
+    def test_window_snapshots_and_never_allocates(self):
+        """Grad buffers are reused across steps; the ring must snapshot
+        into preallocated address-stable rows (zero per-step allocation:
+        VERDICT r1 weak #4, enables hipGraph capture of GNS jobs)."""
         model = torch.nn.Linear(4, 2)
         est = GNSEstimator(model, batch_size=8, window=2)
         for p in model.parameters():
             p.grad = torch.ones_like(p)
         est.on_step()
-        first = est._grads[0].clone()
+        newest = est._rows[-1].clone()
+        addrs = [r.data_ptr() for r in est._rows]
         for p in model.parameters():
             p.grad.fill_(99.0)
-        assert torch.equal(est._grads[0], first)
+        # ring content is a snapshot, not a view of the live grads
+        assert torch.equal(est._rows[-1], newest)
+        est.on_step()
+        # after the shift, the old snapshot moved to row 0 unchanged
+        assert torch.equal(est._rows[0], newest)
+        assert est._rows[-1].abs().max().item() == 99.0
+        # addresses are stable across steps (graph-capturable)
+        assert [r.data_ptr() for r in est._rows] == addrs
+
+    def test_matches_dense_reference_across_steps(self):
+        """Ring-buffer stats equal a naive deque-of-clones reference."""
+        torch.manual_seed(0)
+        model = torch.nn.Linear(6, 3)
+        est = GNSEstimator(model, batch_size=4, window=3)
+        from collections import deque
+        ref_win = deque(maxlen=3)
+        for step in range(6):
+            for p in model.parameters():
+                p.grad = torch.randn_like(p)
+            flat = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+            ref_win.append(flat.clone())
+            est.on_step()
+            if len(ref_win) == 3:
+                mean = torch.stack(list(ref_win)).mean(0)
+                big_sq = (mean * mean).sum()
+                small_sq = (ref_win[-1] ** 2).sum()
+                b_s, b_b = 4.0, 12.0
+                g2 = (b_b * big_sq - b_s * small_sq) / (b_b - b_s)
+                s = (small_sq - big_sq) / (1 / b_s - 1 / b_b)
+                if step == 2:
+                    exp = torch.stack([g2, s])
+                else:
+                    exp = prev * 0.9 + 0.1 * torch.stack([g2, s])
+                prev = exp
+                torch.testing.assert_close(est._avg, exp, rtol=1e-5, atol=1e-6)
 
 
 class TestFusedAdamStateDict:

@@ -176,3 +176,54 @@ class TestTrainingParity:
             torch.testing.assert_close(
                 p1[n], p2[n], rtol=1e-4, atol=1e-6, msg=lambda m: f"{n}: {m}"
             )
+
+
+class TestGNSGraphCapture:
+    def test_gns_ring_capturable_and_matches_eager(self):
+        """the ring-buffer GNS estimator must be
+        hipGraph-capturable (zero per-step allocation) and replays must
+        produce the same EMA as the eager estimator fed the same grads.
+        (Stream capture records without executing, so the capture pass
+        itself has no memory side effects; Python-side branches are in
+        steady state after window+1 warmup steps.)"""
+        from shockwave_amd.adapt import GNSEstimator
+
+        torch.manual_seed(11)
+        model = torch.nn.Linear(257, 63).to(DEV)
+        steps = 8
+        grads = [
+            [torch.randn_like(p) for p in model.parameters()]
+            for _ in range(steps)
+        ]
+
+        # eager reference over all steps
+        est_ref = GNSEstimator(model, batch_size=8, window=2)
+        for s in range(steps):
+            for p, g in zip(model.parameters(), grads[s]):
+                p.grad = g.clone()
+            est_ref.on_step()
+        torch.cuda.synchronize()
+        ref_avg = est_ref._avg.clone()
+
+        # graphed: stable grad staging buffers; 3 eager warmup steps
+        # (= window+1, so ring is full and the EMA branch is steady),
+        # capture once, then drive the remaining steps by replay only
+        est = GNSEstimator(model, batch_size=8, window=2)
+        stage = [torch.zeros_like(p) for p in model.parameters()]
+        for p, st in zip(model.parameters(), stage):
+            p.grad = st
+        warm = 3
+        for s in range(warm):
+            for st, g in zip(stage, grads[s]):
+                st.copy_(g)
+            est.on_step()
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            est.on_step()
+        for s in range(warm, steps):
+            for st, g in zip(stage, grads[s]):
+                st.copy_(g)
+            graph.replay()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(est._avg, ref_avg, rtol=1e-4, atol=1e-6)
