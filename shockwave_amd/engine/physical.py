@@ -64,12 +64,17 @@ class PhysicalScheduler(RoundScheduler):
         port: int = 50070,
         expected_num_workers: int = None,
         completion_buffer_s: float = JOB_COMPLETION_BUFFER_TIME,
+        heartbeat_timeout_s: float = 90.0,
         **kwargs,
     ):
         super().__init__(policy, simulate=False, **kwargs)
         self._port = port
         self._expected_num_workers = expected_num_workers
         self._completion_buffer_s = completion_buffer_s
+        # worker liveness: deregister after this long without a heartbeat
+        # (workers beat every ~30 s); None disables the check
+        self._heartbeat_timeout_s = heartbeat_timeout_s
+        self._last_heartbeat = {}
 
         self._scheduler_lock = threading.RLock()
         self._scheduler_cv = threading.Condition(self._scheduler_lock)
@@ -103,7 +108,7 @@ class PhysicalScheduler(RoundScheduler):
             port,
             {
                 "RegisterWorker": self._register_worker_callback,
-                "SendHeartbeat": lambda: None,
+                "SendHeartbeat": self._heartbeat_callback,
                 "Done": self._done_callback,
                 "InitJob": self._init_job_callback,
                 "UpdateLease": self._update_lease_callback,
@@ -123,11 +128,56 @@ class PhysicalScheduler(RoundScheduler):
         with self._scheduler_cv:
             rpc_client = SchedulerRpcClient(ip_addr, port)
             worker_ids = self.register_worker(worker_type, num_gpus=num_gpus)
+            now = time.time()
             for worker_id in worker_ids:
                 self._worker_connections[worker_id] = rpc_client
                 self._worker_addrs[worker_id] = (ip_addr, port)
+                self._last_heartbeat[worker_id] = now
             self._scheduler_cv.notify_all()
             return worker_ids, self._time_per_iteration
+
+    # ------------------------------------------------------------------
+    # worker liveness (exceeds the reference: its SendHeartbeat is a no-op
+    # and a dead worker's jobs churn through watchdog kills forever)
+    # ------------------------------------------------------------------
+
+    def _heartbeat_callback(self, worker_ids):
+        now = time.time()
+        with self._scheduler_lock:
+            for wid in worker_ids:
+                if wid in self._worker_connections:
+                    self._last_heartbeat[wid] = now
+
+    def _check_worker_liveness(self):
+        """Called at round start (lock held): deregister workers whose
+        heartbeats stopped; their in-flight jobs fall to the watchdog
+        (synthesized zero-step done) and are rescheduled elsewhere."""
+        if self._heartbeat_timeout_s is None:
+            return
+        now = time.time()
+        for wid in list(self._worker_connections):
+            last = self._last_heartbeat.get(wid)
+            if last is not None and now - last > self._heartbeat_timeout_s:
+                self._deregister_worker(wid)
+
+    def _deregister_worker(self, worker_id):
+        logger.error(
+            "worker %d missed heartbeats for >%ss; deregistering",
+            worker_id, self._heartbeat_timeout_s,
+        )
+        if not self.deregister_worker(worker_id):
+            return
+        self._worker_connections.pop(worker_id, None)
+        self._worker_addrs.pop(worker_id, None)
+        self._last_heartbeat.pop(worker_id, None)
+        try:
+            self._available_worker_ids.get_nowait(item=worker_id)
+        except Exception:
+            pass
+        if self.is_shockwave:
+            self._shockwave_planner.ngpus = max(
+                1, sum(self._cluster_spec.values())
+            )
 
     def add_job(self, job, timestamp=None):
         with self._scheduler_cv:
@@ -196,6 +246,7 @@ class PhysicalScheduler(RoundScheduler):
         self._shutdown_event.set()
 
     def _begin_round(self):
+        self._check_worker_liveness()
         self._current_round_start_time = self.get_current_timestamp()
         for job_id in self._current_worker_assignments:
             for single in job_id.singletons():
@@ -280,7 +331,8 @@ class PhysicalScheduler(RoundScheduler):
         for job_id in list(self._jobs_with_extended_lease):
             if job_id in self._jobs:
                 for worker_id in self._current_worker_assignments[job_id]:
-                    self._available_worker_ids.put(worker_id)
+                    if worker_id in self._worker_connections:
+                        self._available_worker_ids.put(worker_id)
             self._jobs_with_extended_lease.discard(job_id)
 
         if not is_final_round:
@@ -353,9 +405,21 @@ class PhysicalScheduler(RoundScheduler):
                         "mps_thread_percentage": job.mps_thread_percentage,
                     }
                 )
-            self._worker_connections[worker_id].run_job(
-                job_descriptions, worker_id, current_round
-            )
+            conn = self._worker_connections.get(worker_id)
+            if conn is None:
+                logger.warning(
+                    "not dispatching %s to deregistered worker %d",
+                    job_id, worker_id,
+                )
+                continue
+            try:
+                conn.run_job(job_descriptions, worker_id, current_round)
+            except Exception:
+                logger.warning(
+                    "RunJob RPC to worker %d failed", worker_id,
+                    exc_info=True,
+                )
+                continue
             if not next_round:
                 try:
                     self._available_worker_ids.get_nowait(item=worker_id)
@@ -395,7 +459,7 @@ class PhysicalScheduler(RoundScheduler):
     def _init_job_callback(self, job_id):
         with self._scheduler_cv:
             if job_id not in self._jobs:
-                return (0, 0, 0)
+                return (0, 0, 0, 0, 0)
             # wait while the job is dispatched for the NEXT round but its
             # workers are still running the current round
             while True:
@@ -420,7 +484,7 @@ class PhysicalScheduler(RoundScheduler):
                 if currently_active and next_combination is not None:
                     self._scheduler_cv.wait(timeout=5)
                     if self._shutdown_event.is_set():
-                        return (0, 0, 0)
+                        return (0, 0, 0, 0, 0)
                 else:
                     break
 
@@ -434,9 +498,17 @@ class PhysicalScheduler(RoundScheduler):
             remaining_steps = int(
                 math.ceil(self._get_remaining_steps(job_id) / scale_factor)
             )
+            # real run_time_so_far + deadline in the InitJob response
+            # (reference UpdateLeaseResponse; r1 stubbed these to 0)
+            run_time_so_far = int(
+                sum(self._cumulative_run_time.get(job_id, {}).values())
+                / scale_factor
+            )
+            deadline = int(self._jobs[job_id].duration * 1.5)
             now = self.get_current_timestamp()
             if self._current_round_start_time is None:
-                return (remaining_steps, self._time_per_iteration, 0)
+                return (remaining_steps, self._time_per_iteration, 0,
+                        run_time_so_far, deadline)
             round_end = (
                 self._current_round_start_time + self._time_per_iteration
             )
@@ -448,13 +520,16 @@ class PhysicalScheduler(RoundScheduler):
             ):
                 # early dispatch for next round: full round + extra time
                 return (remaining_steps, self._time_per_iteration,
-                        remaining_time)
+                        remaining_time, run_time_so_far, deadline)
             if remaining_time > 0:
-                return (remaining_steps, remaining_time, 0)
+                return (remaining_steps, remaining_time, 0,
+                        run_time_so_far, deadline)
             return (
                 remaining_steps,
                 self._time_per_iteration - EARLY_INIT_THRESHOLD,
                 remaining_time,
+                run_time_so_far,
+                deadline,
             )
 
     def _update_lease_callback(
@@ -558,11 +633,19 @@ class PhysicalScheduler(RoundScheduler):
             worker_ids = self._current_worker_assignments[job_id]
             servers = set()
             for worker_id in worker_ids:
-                client = self._worker_connections[worker_id]
+                client = self._worker_connections.get(worker_id)
+                if client is None:
+                    continue  # worker deregistered (liveness)
                 key = (client.addr, client.port)
                 if key not in servers:
                     for single in job_id.singletons():
-                        client.kill_job(single[0])
+                        try:
+                            client.kill_job(single[0])
+                        except Exception:
+                            logger.warning(
+                                "kill RPC to worker %d failed (dead?)",
+                                worker_id, exc_info=True,
+                            )
                     servers.add(key)
             self._completion_events.pop(job_id, None)
 
@@ -668,7 +751,8 @@ class PhysicalScheduler(RoundScheduler):
                 np.max(all_execution_times)
             )
 
-            self._available_worker_ids.put(worker_id)
+            if worker_id in self._worker_connections:
+                self._available_worker_ids.put(worker_id)
             scale_factor = len(self._current_worker_assignments[job_id])
             self._in_progress_updates[job_id].append(
                 (worker_id, all_num_steps, all_execution_times,
@@ -728,12 +812,19 @@ class PhysicalScheduler(RoundScheduler):
             if not micro_task_succeeded:
                 logger.info("[Micro-task failed] job %s", job_id)
                 if not job_id.is_pair() and is_active[job_id]:
-                    self._num_failures_per_job[job_id] += 1
-                    if (
-                        self._num_failures_per_job[job_id]
-                        >= MAX_FAILED_ATTEMPTS
-                    ):
+                    if is_over_deadline:
+                        # zero-step report from a job already past its
+                        # deadline (e.g. the iterator self-completed at
+                        # init): force-complete instead of counting a
+                        # failure (reference deadline handling :4392-4401)
                         to_remove.append(job_id)
+                    else:
+                        self._num_failures_per_job[job_id] += 1
+                        if (
+                            self._num_failures_per_job[job_id]
+                            >= MAX_FAILED_ATTEMPTS
+                        ):
+                            to_remove.append(job_id)
                 self._need_to_update_allocation = True
             else:
                 self._num_failures_per_job[job_id] = 0

@@ -255,6 +255,22 @@ class RoundScheduler:
         self._need_to_update_allocation = True
         return server_worker_ids
 
+    def deregister_worker(self, worker_id: int) -> bool:
+        """Remove a dead worker from the schedulable pool (liveness path —
+        the reference never deregisters: scheduler_server.py:36-99, its
+        SendHeartbeat is a no-op).  The id->type mapping entry is kept so
+        late Done callbacks for in-flight rounds still resolve."""
+        wt = self._worker_id_to_worker_type_mapping.get(worker_id)
+        if wt is None or worker_id not in self._worker_ids:
+            return False
+        self._worker_ids.remove(worker_id)
+        self._cluster_spec[wt] -= 1
+        for server_ids in self._worker_type_to_worker_id_mapping[wt]:
+            if worker_id in server_ids:
+                server_ids.remove(worker_id)
+        self._need_to_update_allocation = True
+        return True
+
     # ------------------------------------------------------------------
     # Job lifecycle
     # ------------------------------------------------------------------
@@ -786,7 +802,12 @@ class RoundScheduler:
                         continue
                     if prev_worker_types.get(job_id) == worker_type:
                         prev_ids = self._current_worker_assignments[job_id]
-                        if all(w not in assigned for w in prev_ids):
+                        # stickiness only if every previous worker is
+                        # still registered (liveness may have removed one)
+                        if all(
+                            w not in assigned and w in self._worker_ids
+                            for w in prev_ids
+                        ):
                             new_assignments[job_id] = prev_ids
                             assigned.update(prev_ids)
                 for job_id, sf in scheduled_jobs[worker_type]:

@@ -48,7 +48,16 @@ def serve_scheduler(port: int, callbacks: Dict[str, Callable]):
                     "round_duration": 0, "error_message": str(e)}
 
     def send_heartbeat(req):
-        callbacks["SendHeartbeat"]()
+        import inspect
+
+        fn = callbacks["SendHeartbeat"]
+        # liveness: the heartbeat carries the sender's worker ids so the
+        # scheduler can track per-worker last-seen times (exceeds the
+        # reference, whose heartbeat is a no-op: scheduler_server.py:36-99)
+        if inspect.signature(fn).parameters:
+            fn(req.get("worker_ids", []))
+        else:
+            fn()
         return {}
 
     def done(req):
@@ -68,13 +77,21 @@ def serve_scheduler(port: int, callbacks: Dict[str, Callable]):
 
     def init_job(req):
         job_id = JobIdPair(req["job_id"], None)
-        max_steps, max_duration, extra_time = callbacks["InitJob"](job_id=job_id)
+        resp = callbacks["InitJob"](job_id=job_id)
+        # callbacks may return the full 5-field UpdateLeaseResponse
+        # (max_steps, max_duration, extra_time, run_time_so_far, deadline
+        # — reference iterator_to_scheduler.proto) or the legacy 3-tuple
+        if len(resp) == 5:
+            max_steps, max_duration, extra_time, run_time, deadline = resp
+        else:
+            max_steps, max_duration, extra_time = resp
+            run_time, deadline = 0, 0
         return {
             "max_steps": max_steps,
             "max_duration": max_duration,
             "extra_time": extra_time,
-            "run_time_so_far": 0,
-            "deadline": 0,
+            "run_time_so_far": run_time,
+            "deadline": deadline,
         }
 
     def update_lease(req):
@@ -229,8 +246,11 @@ class WorkerRpcClient:
             return None, None, resp.get("error_message", "unknown error")
         return resp["worker_ids"], resp["round_duration"], None
 
-    def send_heartbeat(self):
-        self._client.call("WorkerToScheduler", "SendHeartbeat", {})
+    def send_heartbeat(self, worker_ids=None):
+        self._client.call(
+            "WorkerToScheduler", "SendHeartbeat",
+            {"worker_ids": list(worker_ids or [])},
+        )
 
     def notify_scheduler(self, worker_id, job_descriptions):
         """job_descriptions: [(job_id, num_steps, execution_time, log)]"""
@@ -267,7 +287,13 @@ class IteratorRpcClient:
             "IteratorToScheduler", "InitJob", {"job_id": self._job_id},
             timeout=60,
         )
-        return resp["max_steps"], resp["max_duration"], resp["extra_time"]
+        return (
+            resp["max_steps"],
+            resp["max_duration"],
+            resp["extra_time"],
+            resp.get("run_time_so_far", 0),
+            resp.get("deadline", 0),
+        )
 
     def update_lease(self, steps, duration, max_steps, max_duration):
         resp = self._client.call(

@@ -46,7 +46,7 @@ class NullLeaseClient:
     """Infinite lease; used for standalone runs and unit tests."""
 
     def init(self):
-        return INFINITY, INFINITY, 0
+        return INFINITY, INFINITY, 0, 0, INFINITY
 
     def update_lease(self, steps, duration, max_steps, max_duration):
         return INFINITY, INFINITY, 0, INFINITY
@@ -256,7 +256,22 @@ class LeaseIterator:
 
     def _update_lease(self, init=False):
         if init:
-            max_steps, max_duration, extra_time = self._client.init()
+            resp = self._client.init()
+            # InitJob carries the full UpdateLeaseResponse (reference
+            # iterator_to_scheduler.proto): run_time_so_far + deadline let
+            # a restarted job self-complete immediately when it is already
+            # over its deadline (legacy 3-tuple clients still accepted)
+            if len(resp) == 5:
+                max_steps, max_duration, extra_time, run_time, deadline = resp
+                if deadline and run_time > deadline:
+                    self._logger.info(
+                        "run time already exceeds deadline at init",
+                        extra={"event": "LEASE", "status": "DEADLINE"},
+                    )
+                    self.complete(timeout=True)
+                    max_steps, max_duration = 0, 0
+            else:
+                max_steps, max_duration, extra_time = resp
         else:
             (max_steps, max_duration, run_time_so_far, deadline) = (
                 self._client.update_lease(

@@ -34,7 +34,9 @@ class Worker:
         gns_run_dir: str = None,
         data_dir: str = None,
         checkpoint_dir: str = "/tmp/swq_checkpoints",
+        heartbeat_interval_s: float = 30.0,
     ):
+        self._heartbeat_interval_s = heartbeat_interval_s
         if num_gpus is None:
             num_gpus = get_num_gpus()
         assert num_gpus > 0, "no GPUs found"
@@ -96,10 +98,10 @@ class Worker:
         )
         self._heartbeat_thread.start()
 
-    def _heartbeat_loop(self, interval_s: float = 30.0):
-        while not self._done.wait(interval_s):
+    def _heartbeat_loop(self):
+        while not self._done.wait(self._heartbeat_interval_s):
             try:
-                self._rpc_client.send_heartbeat()
+                self._rpc_client.send_heartbeat(self.worker_ids)
             except Exception:
                 logger.warning("heartbeat to scheduler failed", exc_info=True)
 

@@ -47,14 +47,15 @@ def add_job(s, throughputs, sf=1, steps=10000):
 
 class TestInitLease:
     def test_unknown_job(self, sched):
-        assert sched._init_job_callback(JobIdPair(42)) == (0, 0, 0)
+        assert sched._init_job_callback(JobIdPair(42)) == (0, 0, 0, 0, 0)
 
     def test_mid_round_lease_is_remaining_time(self, sched, throughputs):
         jid, _ = add_job(sched, throughputs)
         sched._current_worker_assignments = OrderedDict({jid: (0,)})
         sched._current_round_start_time = sched.get_current_timestamp() - 40
-        steps, duration, extra = sched._init_job_callback(jid)
+        steps, duration, extra, rts, deadline = sched._init_job_callback(jid)
         assert steps == 10000
+        assert rts == 0 and deadline == int(3600 * 1.5)
         assert 55 <= duration <= 60  # 100s round, 40s elapsed
         assert extra == 0
 
@@ -66,7 +67,7 @@ class TestInitLease:
         sched._current_round_start_time = sched.get_current_timestamp() - 70
         sched._next_worker_assignments = OrderedDict({jid: (0,)})
         sched._current_worker_assignments = OrderedDict()
-        steps, duration, extra = sched._init_job_callback(jid)
+        steps, duration, extra, _, _ = sched._init_job_callback(jid)
         assert duration == 100  # a full round
         assert 25 <= extra <= 30
 
@@ -75,14 +76,14 @@ class TestInitLease:
         sched._current_round_start_time = sched.get_current_timestamp() - 101
         sched._next_worker_assignments = None
         sched._current_worker_assignments = OrderedDict()
-        steps, duration, extra = sched._init_job_callback(jid)
+        steps, duration, extra, _, _ = sched._init_job_callback(jid)
         assert duration == 100 - EARLY_INIT_THRESHOLD
 
     def test_multi_gpu_steps_divided(self, sched, throughputs):
         jid, _ = add_job(sched, throughputs, sf=4, steps=1000)
         sched._current_worker_assignments = OrderedDict({jid: (0, 1, 2, 3)})
         sched._current_round_start_time = sched.get_current_timestamp()
-        steps, _, _ = sched._init_job_callback(jid)
+        steps = sched._init_job_callback(jid)[0]
         assert steps == 250
 
 

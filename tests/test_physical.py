@@ -357,3 +357,103 @@ class TestMultiWorkerCluster:
             assert any(len(set(w)) == 2 for w in dp_rounds)
         finally:
             sched.shutdown()
+
+
+class TestWorkerLiveness:
+    def test_silent_worker_deregistered(self, throughputs):
+        """Unit-level liveness: a worker that stops heartbeating is
+        deregistered at the next round boundary; live workers stay."""
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+
+        sched = PhysicalScheduler(
+            get_policy("max_min_fairness"),
+            port=free_port(),
+            expected_num_workers=99,  # round loop stays parked
+            throughputs=throughputs,
+            time_per_iteration=100,
+            profiles=[],
+            worker_type="mi355x",
+            heartbeat_timeout_s=0.2,
+        )
+        try:
+            a = sched._register_worker_callback("mi355x", 1, "127.0.0.1", 1)
+            b = sched._register_worker_callback("mi355x", 1, "127.0.0.1", 2)
+            wa, wb = a[0][0], b[0][0]
+            assert sched._cluster_spec["mi355x"] == 2
+            time.sleep(0.3)
+            sched._heartbeat_callback([wa])  # only A beats
+            with sched._scheduler_cv:
+                sched._check_worker_liveness()
+            assert wb not in sched._worker_connections
+            assert wa in sched._worker_connections
+            assert sched._cluster_spec["mi355x"] == 1
+            assert wb not in sched._worker_ids
+            # id->type mapping survives for late Done callbacks
+            assert sched._worker_id_to_worker_type_mapping[wb] == "mi355x"
+        finally:
+            sched.shutdown(shutdown_workers=False)
+
+
+@pytest.mark.slow
+class TestWorkerDeathMidTrace:
+    def test_jobs_complete_after_worker_dies(self, tmp_path, throughputs):
+        """kill a worker mid-trace; its job must be
+        rescheduled onto the survivor and all jobs must complete."""
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        port = free_port()
+        jobs = [tiny_job(steps=8), tiny_job(steps=8)]
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+        sched = PhysicalScheduler(
+            get_policy("max_min_fairness"),
+            port=port,
+            expected_num_workers=2,
+            throughputs=throughputs,
+            time_per_iteration=25,
+            profiles=profiles,
+            worker_type="mi355x",
+            heartbeat_timeout_s=8.0,
+        )
+        workers = []
+        try:
+            for _ in range(2):
+                workers.append(Worker(
+                    worker_type="mi355x",
+                    sched_addr="127.0.0.1",
+                    sched_port=port,
+                    worker_port=free_port(),
+                    num_gpus=1,
+                    ip_addr="127.0.0.1",
+                    run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                    checkpoint_dir=str(tmp_path),
+                    heartbeat_interval_s=2.0,
+                ))
+            for j in jobs:
+                sched.add_job(j)
+            # let round 0 dispatch both jobs (one per worker), then kill
+            # worker B: stop heartbeats + server, kill its job processes
+            time.sleep(6)
+            dead = workers[1]
+            dead._done.set()
+            dead._server.stop(0)
+            dead._dispatcher.shutdown()
+
+            deadline = time.time() + 300
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(2)
+            completions = sched.get_job_completion_times()
+            assert len(completions) == 2, (
+                f"jobs did not complete after worker death: {completions}"
+            )
+            # the dead worker must have been deregistered
+            assert len(sched._worker_connections) == 1
+        finally:
+            sched.shutdown()
+            for w in workers[:1]:
+                try:
+                    w._shutdown_callback()
+                except Exception:
+                    pass

@@ -65,7 +65,7 @@ class TestSchedulerRpcLoopback:
 
         def init_job(job_id):
             calls["init"] = job_id
-            return 500, 60.0, 3.0
+            return 500, 60.0, 3.0, 17, 9000
 
         def update_lease(job_id, worker_id, steps, duration, max_steps,
                          max_duration):
@@ -79,7 +79,8 @@ class TestSchedulerRpcLoopback:
             port,
             {
                 "RegisterWorker": register,
-                "SendHeartbeat": lambda: None,
+                "SendHeartbeat": lambda wids: calls.setdefault(
+                    "heartbeats", []).append(list(wids)),
                 "Done": done,
                 "InitJob": init_job,
                 "UpdateLease": update_lease,
@@ -96,10 +97,12 @@ class TestSchedulerRpcLoopback:
             assert calls["done"] == (JobIdPair(3), 0, [100])
 
             ic = IteratorRpcClient(7, 0, "127.0.0.1", port)
-            assert ic.init() == (500, 60.0, 3.0)
+            assert ic.init() == (500, 60.0, 3.0, 17, 9000)
             assert ic.update_lease(10, 5.0, 500, 60.0) == (1000, 120.0, 42, 10000)
             ic.update_resource_requirement(True, False)
             assert calls["rr"] == (JobIdPair(7), True, False)
+            wc.send_heartbeat([0, 1])
+            assert calls["heartbeats"] == [[0, 1]]
         finally:
             server.stop(0)
 
