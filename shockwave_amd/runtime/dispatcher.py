@@ -28,6 +28,7 @@ import re
 import signal
 import subprocess
 import threading
+import time
 from concurrent.futures import ThreadPoolExecutor
 from typing import Dict, List, Optional
 
@@ -54,6 +55,7 @@ class Dispatcher:
         static_run_dir: Optional[str] = None,
         accordion_run_dir: Optional[str] = None,
         gns_run_dir: Optional[str] = None,
+        warm: bool = True,
     ):
         self._round_duration = round_duration
         self._gpu_ids = gpu_ids
@@ -75,6 +77,18 @@ class Dispatcher:
         self._procs: Dict[int, subprocess.Popen] = {}  # job_id -> proc
         self._killed = set()
         self._pool = ThreadPoolExecutor(max_workers=max(8, 2 * len(gpu_ids)))
+        # persistent warm runners (one+ per GPU slot): jobs run in-process
+        # in a long-lived python that holds torch/HIP/MIOpen state, so a
+        # dispatch costs ~0.1 s instead of full process startup
+        self._warm = warm
+        self._runner_pool = None
+        self._job_runners: Dict[int, object] = {}  # job_id -> WarmRunner
+        if warm:
+            from .warm_runner import RunnerPool
+
+            self._runner_pool = RunnerPool(
+                os.path.join(checkpoint_dir, ".runners")
+            )
 
     # -- command construction ----------------------------------------------
 
@@ -126,22 +140,69 @@ class Dispatcher:
 
     # -- launch / kill -------------------------------------------------------
 
+    def _job_env(self, job, worker_id, round_id, gpu_id, pin_gpu=True):
+        env = {
+            "GAVEL_JOB_ID": str(job["job_id"]),
+            "GAVEL_WORKER_ID": str(worker_id),
+            "GAVEL_ROUND_ID": str(round_id),
+            "GAVEL_SCHED_ADDR": self._sched_addr,
+            "GAVEL_SCHED_PORT": str(self._sched_port),
+            "SWQ_MODE": job.get("mode", "static"),
+        }
+        if pin_gpu:
+            env.update(
+                {
+                    "HIP_VISIBLE_DEVICES": str(gpu_id),
+                    "ROCR_VISIBLE_DEVICES": str(gpu_id),
+                    "CUDA_VISIBLE_DEVICES": str(gpu_id),
+                }
+            )
+        return env
+
+    def launch_job_warm(self, job, command, worker_id, round_id, gpu_id):
+        """Run a lease inside the GPU slot's persistent runner."""
+        job_id = job["job_id"]
+        runner = self._runner_pool.acquire(gpu_id)
+        cwd = self._job_working_dir(job)
+        # runner already pins the GPU through its own env
+        env = self._job_env(job, worker_id, round_id, gpu_id, pin_gpu=False)
+        log_path = os.path.join(
+            self._checkpoint_dir, f"job_id={job_id}", "job_output.log"
+        )
+        logger.info(
+            "[worker %s round %s] warm-dispatching job %s on gpu %s: %s",
+            worker_id, round_id, job_id, gpu_id, command,
+        )
+        with self._lock:
+            self._job_runners[job_id] = runner
+        try:
+            rc = runner.run(command, cwd, env, log_path)
+        finally:
+            with self._lock:
+                self._job_runners.pop(job_id, None)
+                was_killed = job_id in self._killed
+                self._killed.discard(job_id)
+            self._runner_pool.release(runner)
+        if rc != 0 and not was_killed:
+            tail = ""
+            try:
+                with open(log_path, "rb") as f:
+                    tail = b"\n".join(
+                        f.read().splitlines()[-20:]
+                    ).decode(errors="replace")
+            except OSError:
+                pass
+            logger.error("job %s exited rc=%s; output tail:\n%s",
+                         job_id, rc, tail)
+        return rc, b""
+
     def launch_job(self, job, command, worker_id, round_id, gpu_id):
+        if self._warm:
+            return self.launch_job_warm(job, command, worker_id, round_id,
+                                        gpu_id)
         job_id = job["job_id"]
         env = dict(os.environ)
-        env.update(
-            {
-                "GAVEL_JOB_ID": str(job_id),
-                "GAVEL_WORKER_ID": str(worker_id),
-                "GAVEL_ROUND_ID": str(round_id),
-                "GAVEL_SCHED_ADDR": self._sched_addr,
-                "GAVEL_SCHED_PORT": str(self._sched_port),
-                "HIP_VISIBLE_DEVICES": str(gpu_id),
-                "ROCR_VISIBLE_DEVICES": str(gpu_id),
-                "CUDA_VISIBLE_DEVICES": str(gpu_id),
-                "SWQ_MODE": job.get("mode", "static"),
-            }
-        )
+        env.update(self._job_env(job, worker_id, round_id, gpu_id))
         cwd = self._job_working_dir(job)
         logger.info(
             "[worker %s round %s] launching job %s on gpu %s: %s (cwd %s)",
@@ -227,8 +288,26 @@ class Dispatcher:
     def kill_job(self, job_id):
         with self._lock:
             proc = self._procs.get(job_id)
-            if proc is not None:
+            runner = self._job_runners.get(job_id)
+            if proc is not None or runner is not None:
                 self._killed.add(job_id)
+        if runner is not None:
+            # SIGTERM the runner: the workload loop's graceful handler
+            # checkpoints and returns (runner survives); if the job is
+            # hung the runner gets SIGKILLed and replaced
+            logger.info("killing warm job %s (runner pid %s)",
+                        job_id, runner.proc.pid)
+            runner.terminate()
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                with self._lock:
+                    if job_id not in self._job_runners:
+                        return
+                time.sleep(0.25)
+            logger.warning("job %s did not stop in 10 s; killing runner",
+                           job_id)
+            runner.kill()
+            return
         if proc is None:
             logger.info("kill_job(%s): no running process", job_id)
             return
@@ -244,10 +323,12 @@ class Dispatcher:
 
     def reset(self):
         with self._lock:
-            job_ids = list(self._procs.keys())
+            job_ids = list(self._procs.keys()) + list(self._job_runners.keys())
         for job_id in job_ids:
             self.kill_job(job_id)
 
     def shutdown(self):
         self.reset()
+        if self._runner_pool is not None:
+            self._runner_pool.shutdown()
         self._pool.shutdown(wait=False)

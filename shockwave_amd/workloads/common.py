@@ -44,8 +44,12 @@ def init_device_and_distributed(args) -> torch.device:
     """Bind the GPU and bring up RCCL (gloo on CPU) when distributed."""
     use_cuda = torch.cuda.is_available()
     if use_cuda:
-        torch.cuda.set_device(args.local_rank)
-        device = torch.device("cuda", args.local_rank)
+        # the dispatcher both pins HIP_VISIBLE_DEVICES to this job's GPU
+        # and passes --local_rank <gpu>: inside the process the pinned
+        # GPU is device 0, so clamp to the visible range
+        local = min(args.local_rank, torch.cuda.device_count() - 1)
+        torch.cuda.set_device(local)
+        device = torch.device("cuda", local)
     else:
         device = torch.device("cpu")
     if args.world_size > 1 and not dist.is_initialized():
