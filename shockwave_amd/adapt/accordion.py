@@ -101,6 +101,14 @@ class AccordionDetector:
         for a, saved in zip(self.accum, state["accum"]):
             a.copy_(saved.to(a.device))
 
+    def reset(self) -> None:
+        """Fresh-job state, in place (session-cache reuse)."""
+        for a in self.accum:
+            a.zero_()
+        self.norms_by_epoch = {}
+        self._prev_checkpoint_sum = None
+        self.in_critical_regime = True
+
 
 def hardcoded_critical_regime(model: str, original_bs: int, epoch: int) -> bool:
     """The hard-coded regime tables the reference uses in practice

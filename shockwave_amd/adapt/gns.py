@@ -129,9 +129,21 @@ class GNSEstimator:
 
     def load_state_dict(self, state):
         if state.get("g2_avg") is not None:
-            self._avg = torch.tensor(
-                [state["g2_avg"], state["s_avg"]], device=self._device
-            )
+            # in place: a session-cached hipGraph references _avg's address
+            self._avg.copy_(torch.tensor(
+                [state["g2_avg"], state["s_avg"]], dtype=self._avg.dtype
+            ))
             self._have_avg = True
         self.gns_by_epoch = {int(k): v for k, v in state["gns_by_epoch"].items()}
         self.batch_size = state["batch_size"]
+
+    def reset(self, batch_size: Optional[int] = None) -> None:
+        """Return to the fresh-job state in place (session-cache reuse by
+        a job with no checkpoint)."""
+        self._ring.zero_()
+        self._filled = 0
+        self._avg.zero_()
+        self._have_avg = False
+        self.gns_by_epoch = {}
+        if batch_size is not None:
+            self.batch_size = batch_size

@@ -18,6 +18,53 @@ import torch
 from . import fused_adam, fused_sgd
 
 
+def _load_state_inplace(opt, state_dict, tensor_keys):
+    """Copy a checkpoint's per-param state tensors INTO the optimizer's
+    existing state tensors (same addresses) instead of replacing them —
+    a session-cached hipGraph holds the old addresses.  Returns False on
+    any structural mismatch (caller falls back to the replacing load)."""
+    groups = opt.param_groups
+    saved_groups = state_dict.get("param_groups")
+    saved_state = state_dict.get("state", {})
+    if saved_groups is None or len(saved_groups) != len(groups):
+        return False
+    # torch state_dict convention: params are indexed in group order
+    own_params = [p for g in groups for p in g["params"]]
+    saved_ids = [i for g in saved_groups for i in g["params"]]
+    if len(own_params) != len(saved_ids):
+        return False
+    for p, sid in zip(own_params, saved_ids):
+        saved = saved_state.get(sid, {})
+        cur = opt.state.get(p, {})
+        for key, val in saved.items():
+            if torch.is_tensor(val) and val.dim() > 0:
+                if key not in cur or cur[key].shape != val.shape:
+                    return False
+        for key, val in saved.items():
+            if torch.is_tensor(val) and val.dim() > 0:
+                cur[key].copy_(val.to(cur[key].device))
+            else:
+                cur[key] = val
+    for g, sg in zip(groups, saved_groups):
+        for k, v in sg.items():
+            if k != "params":
+                g[k] = v
+    return True
+
+
+def reset_optimizer_state_inplace(opt):
+    """Zero all state tensors in place (fresh job reusing a cached
+    session) without changing addresses."""
+    for s in opt.state.values():
+        for v in s.values():
+            if torch.is_tensor(v) and v.dim() > 0:
+                v.zero_()
+    if isinstance(opt, FusedAdam):
+        opt._step_count = 0
+    if isinstance(opt, FusedSGD):
+        opt._buffers_initialized = False
+
+
 def _preinit_grads(param_groups):
     """Create zero gradients with each param's own memory layout before
     the first backward: autograd then accumulates into them, so grad
@@ -81,6 +128,13 @@ class FusedSGD(torch.optim.Optimizer):
         return loss
 
     def load_state_dict(self, state_dict):
+        if _load_state_inplace(self, state_dict, ("momentum_buffer",)):
+            # addresses unchanged: cached lists (and any captured graph)
+            # stay valid
+            self._buffers_initialized = any(
+                "momentum_buffer" in s for s in self.state.values()
+            )
+            return
         super().load_state_dict(state_dict)
         self._cached_lists = None  # state tensors were replaced
         # restored momentum buffers must accumulate, not be overwritten
@@ -156,6 +210,17 @@ class FusedAdam(torch.optim.Optimizer):
                 for s in state_dict.get("state", {}).values()
             ]
             self._step_count = max(steps) if steps else 0
+        # drop a stock-Adam 'step' scalar so the in-place path does not
+        # stash it; our step count is _step_count
+        slim = {
+            "param_groups": state_dict.get("param_groups"),
+            "state": {
+                k: {kk: vv for kk, vv in s.items() if kk != "step"}
+                for k, s in state_dict.get("state", {}).items()
+            },
+        }
+        if _load_state_inplace(self, slim, ("exp_avg", "exp_avg_sq")):
+            return
         super().load_state_dict(state_dict)
         self._cached_lists = None
 

@@ -202,6 +202,15 @@ class LeaseIterator:
         self._write_info()
         self._file_handler.flush()
 
+    def close(self):
+        """Release the log handler and drop the atexit hooks (the hooks
+        are crash insurance; a clean end must not leave one per lease
+        piling up inside a long-lived warm runner)."""
+        atexit.unregister(self._close_file_handler)
+        if self._write_on_close:
+            atexit.unregister(self._write_info)
+        self._close_file_handler()
+
     def complete(self, timeout=False):
         self._done = True
         if not self._write_on_close:
