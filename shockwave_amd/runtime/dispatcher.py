@@ -87,7 +87,13 @@ class Dispatcher:
             from .warm_runner import RunnerPool
 
             self._runner_pool = RunnerPool(
-                os.path.join(checkpoint_dir, ".runners")
+                os.path.join(checkpoint_dir, ".runners"),
+                extra_env={
+                    # set at spawn so MIOpen picks it up at library init
+                    "MIOPEN_USER_DB_PATH": os.path.join(
+                        checkpoint_dir, ".miopen"
+                    ),
+                },
             )
 
     # -- command construction ----------------------------------------------
@@ -148,6 +154,12 @@ class Dispatcher:
             "GAVEL_SCHED_ADDR": self._sched_addr,
             "GAVEL_SCHED_PORT": str(self._sched_port),
             "SWQ_MODE": job.get("mode", "static"),
+            # MIOpen find-db persists under the (typically shared)
+            # checkpoint dir: a migrated job's destination node skips the
+            # convolution-algorithm search its source already paid for
+            "MIOPEN_USER_DB_PATH": os.path.join(
+                self._checkpoint_dir, ".miopen"
+            ),
         }
         if pin_gpu:
             env.update(

@@ -283,10 +283,12 @@ class RunnerPool:
     """Per-GPU pools of warm runners; spawns on demand (a packed pair
     needs two co-resident runners on one GPU)."""
 
-    def __init__(self, runner_log_dir: str):
+    def __init__(self, runner_log_dir: str,
+                 extra_env: Optional[Dict[str, str]] = None):
         self._pools: Dict[int, List[WarmRunner]] = {}
         self._lock = threading.Lock()
         self._log_dir = runner_log_dir
+        self._extra_env = extra_env or {}
         os.makedirs(runner_log_dir, exist_ok=True)
 
     def acquire(self, gpu_id: int) -> WarmRunner:
@@ -301,6 +303,7 @@ class RunnerPool:
         runner = WarmRunner(
             gpu_id,
             os.path.join(self._log_dir, f"runner_gpu{gpu_id}.log"),
+            extra_env=self._extra_env,
         )
         runner.busy = True
         with self._lock:
