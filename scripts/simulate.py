@@ -33,6 +33,8 @@ def run_simulation(
     results_dir=None,
     log_level="WARNING",
     preemption_overhead_s=20.0,
+    warm_overhead_s=None,
+    midround_staleness=False,
     ideal=False,
 ):
     import logging
@@ -81,6 +83,8 @@ def run_simulation(
         shockwave_config=shockwave_config,
         worker_type=worker_type,
         preemption_overhead_s=preemption_overhead_s,
+        warm_preemption_overhead_s=warm_overhead_s,
+        midround_staleness=midround_staleness,
     )
 
     start = time.time()
@@ -152,6 +156,14 @@ def main():
     p.add_argument("--log_level", default="WARNING")
     p.add_argument("--preemption_overhead", type=float, default=20.0,
                    help="simulated checkpoint/restart cost per migration (s)")
+    p.add_argument("--midround_staleness", action="store_true",
+                   help="model the physical planner's mid-round decision "
+                        "point (half of the current round's service "
+                        "unobserved) for non-shockwave policies")
+    p.add_argument("--warm_overhead", type=float, default=None,
+                   help="migration cost once the job type has run before "
+                        "(MIOpen find-db / warm-runner session hot); "
+                        "default: flat model")
     args = p.parse_args()
 
     if args.generate_jobs:
@@ -186,6 +198,8 @@ def main():
         results_dir=args.results_dir,
         log_level=args.log_level,
         preemption_overhead_s=args.preemption_overhead,
+        warm_overhead_s=args.warm_overhead,
+        midround_staleness=args.midround_staleness,
         ideal=args.ideal,
     )
     print(
