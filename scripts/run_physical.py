@@ -145,6 +145,25 @@ def main(argv=None):
         "worst_ftf_rho": max(ftf[0]) if ftf and ftf[0] else None,
         "cluster_util": util[0],
         "num_completed": len(sched.get_job_completion_times()),
+        # raw per-job evidence for scripts/analyze_jobs.py: iterator
+        # event logs per (job, worker slot), per-round schedules, and the
+        # per-round (throughput, bs) timeline
+        "job_timelines": {
+            str(jid): logs for jid, logs in sched._job_timelines.items()
+        },
+        "per_round_schedule": [
+            {str(j): list(w) for j, w in rnd.items()}
+            for rnd in sched._per_round_schedule
+        ],
+        "throughput_timeline": {
+            int(j): dict(t) for j, t in sched._throughput_timeline.items()
+        },
+        "job_total_steps": {
+            str(jid): job.total_steps for jid, job in sched._jobs.items()
+        },
+        "completion_times": {
+            str(j): t for j, t in sched.get_job_completion_times().items()
+        },
     }
     sched.shutdown()
     print(json.dumps(results))
