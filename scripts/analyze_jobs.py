@@ -55,17 +55,25 @@ def parse_timeline(lines):
 
 
 def split_dispatches(events):
-    """Split a slot's event stream into dispatches (INIT ... last event)."""
+    """Split a slot's event stream into dispatches.  Each lease starts by
+    logging PROGRESS STEPS 0; a drop in the cumulative step counter marks
+    the next dispatch."""
     dispatches = []
     cur = None
+    last_steps = -1
     for ev in events:
-        if ev["event"] == "INIT" or (
-            cur is None and ev["event"] == "LEASE"
-        ):
+        is_boundary = ev["event"] == "INIT"
+        if ev["event"] == "PROGRESS" and ev["status"] == "STEPS":
+            try:
+                v = int(float(ev["msg"]))
+            except ValueError:
+                v = last_steps
+            if v < last_steps:
+                is_boundary = True
+            last_steps = v
+        if cur is None or is_boundary:
             if cur:
                 dispatches.append(cur)
-            cur = {"events": []}
-        if cur is None:
             cur = {"events": []}
         cur["events"].append(ev)
     if cur:

@@ -113,13 +113,18 @@ def main(argv=None):
         )
 
     # paced submission (reference run_scheduler_with_trace.py:38-120)
+    job_types = {}
+
     def submit():
         start = time.time()
         for job, at in zip(jobs, arrival_times):
             delay = at * args.arrival_time_scale - (time.time() - start)
             if delay > 0:
                 time.sleep(delay)
-            sched.add_job(job)
+            jid = sched.add_job(job)
+            job_types[str(jid[0] if hasattr(jid, "__getitem__") else jid)] = (
+                job.job_type
+            )
 
     submitter = threading.Thread(target=submit, daemon=True)
     submitter.start()
@@ -158,6 +163,7 @@ def main(argv=None):
         "throughput_timeline": {
             int(j): dict(t) for j, t in sched._throughput_timeline.items()
         },
+        "job_types": job_types,
         "job_total_steps": {
             str(jid): job.total_steps for jid, job in sched._jobs.items()
         },

@@ -35,6 +35,8 @@ def run_simulation(
     preemption_overhead_s=20.0,
     warm_overhead_s=None,
     midround_staleness=False,
+    fixed_rounds=False,
+    startup_table=None,
     ideal=False,
 ):
     import logging
@@ -85,6 +87,8 @@ def run_simulation(
         preemption_overhead_s=preemption_overhead_s,
         warm_preemption_overhead_s=warm_overhead_s,
         midround_staleness=midround_staleness,
+        fixed_rounds=fixed_rounds,
+        startup_table=startup_table,
     )
 
     start = time.time()
@@ -160,6 +164,13 @@ def main():
                    help="model the physical planner's mid-round decision "
                         "point (half of the current round's service "
                         "unobserved) for non-shockwave policies")
+    p.add_argument("--startup_table", default=None,
+                   help="JSON of per-job-type first-dispatch startup "
+                        "seconds (scripts/calibrate_sim.py)")
+    p.add_argument("--fixed_rounds", action="store_true",
+                   help="wall-clock round boundaries (round-tail idle "
+                        "after early completions), as the physical "
+                        "mechanism")
     p.add_argument("--warm_overhead", type=float, default=None,
                    help="migration cost once the job type has run before "
                         "(MIOpen find-db / warm-runner session hot); "
@@ -200,6 +211,9 @@ def main():
         preemption_overhead_s=args.preemption_overhead,
         warm_overhead_s=args.warm_overhead,
         midround_staleness=args.midround_staleness,
+        fixed_rounds=args.fixed_rounds,
+        startup_table=(json.load(open(args.startup_table))
+                       if args.startup_table else None),
         ideal=args.ideal,
     )
     print(
