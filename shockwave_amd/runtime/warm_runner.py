@@ -139,12 +139,22 @@ def _run_one(req, base_env, base_cwd, state):
             or tokens[0] == "-u"
         ):
             tokens.pop(0)
-        script, argv = tokens[0], tokens[1:]
-        script = os.path.join(cwd, script) if not os.path.isabs(script) else script
-        sys.argv = [script] + argv
         state["in_job"] = True
         try:
-            runpy.run_path(script, run_name="__main__")
+            if tokens and tokens[0] == "-c":
+                # `python -c "code"` commands (tests, ad-hoc probes)
+                code, argv = tokens[1], tokens[2:]
+                sys.argv = ["-c"] + argv
+                exec(compile(code, "<command>", "exec"),
+                     {"__name__": "__main__"})
+            else:
+                script, argv = tokens[0], tokens[1:]
+                script = (
+                    os.path.join(cwd, script)
+                    if not os.path.isabs(script) else script
+                )
+                sys.argv = [script] + argv
+                runpy.run_path(script, run_name="__main__")
         except SystemExit as e:
             rc = int(e.code or 0) if not isinstance(e.code, str) else 1
         print(f"[warm_runner] job finished rc={rc}", flush=True)
