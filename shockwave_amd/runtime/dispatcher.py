@@ -209,7 +209,11 @@ class Dispatcher:
         return rc, b""
 
     def launch_job(self, job, command, worker_id, round_id, gpu_id):
-        if self._warm:
+        # the warm runner executes python entrypoints in-process; anything
+        # else (shell pipelines, non-python tools) takes the cold path
+        if self._warm and os.path.basename(
+            command.split()[0]
+        ).startswith("python"):
             return self.launch_job_warm(job, command, worker_id, round_id,
                                         gpu_id)
         job_id = job["job_id"]
