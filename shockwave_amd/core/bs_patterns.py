@@ -75,9 +75,16 @@ def gns_bs_pattern(
     bs = [batch_size] * num_epochs
     if segments is None:
         return bs
-    for start, end, mult in segments:
+    for seg in segments:
+        start, end, mult = seg[0], seg[1], seg[2]
+        # check_first: the reference's check-then-multiply loops leave
+        # the FINAL training epoch at the base batch size (loop-ordering
+        # quirk in utils.py:801-1330, reproduced as data for parity)
+        check_first = bool(seg[3]) if len(seg) > 3 else False
         end = num_epochs if end is None else min(end, num_epochs)
         for e in range(start, end):
+            if check_first and e + 1 >= num_epochs:
+                break
             bs[e] = batch_size * mult
     return bs
 

@@ -258,11 +258,12 @@ class ShockwavePlanner:
                 priorities.append(math.exp(min(log_p, 700.0)))
             else:
                 priorities.append(1.0)
-        # normalize giant priorities to avoid numeric blowup in the MILP
-        arr = np.array(priorities)
-        cap = 1e6
-        if arr.max() > cap:
-            arr = arr / arr.max() * cap
+        # clamp giant priorities to avoid numeric blowup in the MILP.
+        # Clamp PER JOB (min), never rescale the whole vector: dividing
+        # everything by one near-done job's astronomic ratio^100 crushed
+        # moderate violators' boosts below the k*M term and starved them
+        # for dozens of rounds (the r1 worst-rho tail: VERDICT item 3)
+        arr = np.minimum(np.array(priorities), 1e6)
         return arr.tolist()
 
     def _greedy_schedule(self, job_ids) -> np.ndarray:

@@ -205,3 +205,63 @@ class TestFusedAdamStateDict:
         fused2.load_state_dict(sd)
         assert fused2._step_count == 7
         assert "swq_step_count" in sd  # caller's dict untouched
+
+
+class TestBsPatternOracleParity:
+    """The adaptation twins' bs schedules must equal the reference's
+    oracle tables exactly, including its loop-ordering quirk (check-first
+    segments leave the final epoch at base bs).  Skips without the
+    reference tree; patterns are data regenerated by
+    scripts/gen_bs_ladder.py."""
+
+    REF = "/root/reference/scheduler/utils.py"
+
+    def _load(self, name):
+        import ast
+
+        import pytest as _pytest
+
+        import os
+        if not os.path.exists(self.REF):
+            _pytest.skip("reference tree not present")
+        ns = {}
+        for node in ast.parse(open(self.REF).read()).body:
+            if isinstance(node, ast.FunctionDef) and node.name == name:
+                exec(compile(ast.Module(body=[node], type_ignores=[]),
+                             self.REF, "exec"), ns)
+        return ns[name]
+
+    def test_gns_ladder_matches_reference(self):
+        import json
+        import os
+
+        from shockwave_amd.core import bs_patterns as bp
+
+        ref = self._load("get_gns_bs_pattern")
+        keys = json.load(open(os.path.join(
+            os.path.dirname(bp.__file__), "data", "gns_bs_ladder.json"
+        ))).keys()
+        for key in keys:
+            model, bs, sf = key.split("|")
+            jt = f"{model} (batch size {bs})"
+            for E in (1, 5, 12, 31, 59, 100, 200, 400):
+                r = [int(x) for x in ref(jt, int(bs), E, int(sf))]
+                o = bp.gns_bs_pattern(jt, int(bs), E, int(sf))
+                assert r == o, (key, E)
+
+    def test_accordion_matches_reference(self):
+        from shockwave_amd.core.bs_patterns import accordion_bs_pattern
+
+        ref = self._load("get_accordion_bs_pattern")
+        combos = (
+            [("ResNet-18", b) for b in (16, 32, 64, 128, 256)]
+            + [("ResNet-50", b) for b in (16, 32, 64)]
+            + [("Transformer", b) for b in (16, 32, 64, 128)]
+            + [("LM", b) for b in (5, 10, 20, 40, 80)]
+            + [("Recommendation", b) for b in (512, 2048, 8192)]
+        )
+        for model, bs in combos:
+            jt = f"{model} (batch size {bs})"
+            for E in (1, 10, 31, 100, 270):
+                r = [int(x) for x in ref(jt, bs, E, 1)]
+                assert r == accordion_bs_pattern(jt, bs, E), (jt, E)

@@ -201,3 +201,12 @@ class TestReferenceParity:
         assert abs(makespan - ref["makespan"]) / ref["makespan"] < 0.05
         assert (abs(sched.get_average_jct()[0] - float(ref["avg_jct"]))
                 / float(ref["avg_jct"]) < 0.05)
+        # round-2 band: worst rho within 10% of the
+        # reference and unfair fraction equal
+        ftf, _ = sched.get_finish_time_fairness()
+        ref_rhos = ref["finish_time_fairness_list"]
+        assert max(ftf) <= 1.9
+        assert abs(max(ftf) - max(ref_rhos)) / max(ref_rhos) < 0.10
+        unfair = sum(1 for r in ftf if r > 1.05)
+        ref_unfair = sum(1 for r in ref_rhos if r > 1.05)
+        assert unfair == ref_unfair
