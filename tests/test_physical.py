@@ -457,3 +457,88 @@ class TestWorkerDeathMidTrace:
                     w._shutdown_callback()
                 except Exception:
                     pass
+
+
+@pytest.mark.slow
+class TestEightWorkerDressRehearsal:
+    def test_mixed_trace_on_eight_workers(self, tmp_path, throughputs):
+        """8 worker slots (2 daemons x 4 GPUs), a
+        compressed mixed trace — all three modes, scale factors 1/2/4,
+        five families — at short rounds on CPU (gloo).  Shakes out
+        dispatch/port/lease races ahead of the driver's 8-GPU node."""
+        import random
+
+        from shockwave_amd.engine.physical import PhysicalScheduler
+        from shockwave_amd.policies import get_policy
+        from shockwave_amd.runtime.worker import Worker
+
+        rng = random.Random(0)
+        specs = [
+            ("ResNet-18 (batch size 16)",
+             "python3 main.py --batch_size 16",
+             "image_classification/cifar10", "--num_steps"),
+            ("ResNet-18 (batch size 32)",
+             "python3 main.py --batch_size 32",
+             "image_classification/cifar10", "--num_steps"),
+            ("LM (batch size 10)",
+             "python3 main.py --batch_size 10",
+             "language_modeling", "--steps"),
+            ("Recommendation (batch size 512)",
+             "python3 train.py --batch_size 512",
+             "recommendation", "-n"),
+        ]
+        jobs = []
+        for i in range(16):
+            jt, cmd, wd, steps_arg = specs[i % len(specs)]
+            sf = rng.choices([1, 2, 4], weights=[0.6, 0.3, 0.1])[0]
+            mode = rng.choice(["static", "static", "gns", "accordion"])
+            if "Recommendation" in jt:
+                sf = 1  # not distributed in the reference (SURVEY 2.2)
+            jobs.append(Job(
+                job_id=None, job_type=jt, command=cmd,
+                working_directory=wd, num_steps_arg=steps_arg,
+                total_steps=rng.randint(6, 12) * sf,
+                duration=600, scale_factor=sf, mode=mode,
+            ))
+        profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
+
+        port = free_port()
+        sched = PhysicalScheduler(
+            get_policy("max_min_fairness"),
+            port=port,
+            expected_num_workers=8,
+            throughputs=throughputs,
+            time_per_iteration=30,
+            profiles=profiles,
+            worker_type="mi355x",
+            heartbeat_timeout_s=120.0,
+            completion_buffer_s=90.0,
+        )
+        workers = []
+        try:
+            for w in range(2):
+                workers.append(Worker(
+                    worker_type="mi355x",
+                    sched_addr="127.0.0.1",
+                    sched_port=port,
+                    worker_port=free_port(),
+                    num_gpus=4,
+                    ip_addr="127.0.0.1",
+                    run_dir=os.path.join(REPO, "workloads", "pytorch"),
+                    checkpoint_dir=str(tmp_path / f"node{w}"),
+                    heartbeat_interval_s=5.0,
+                ))
+            for j in jobs:
+                sched.add_job(j)
+            deadline = time.time() + 540
+            while not sched.is_done() and time.time() < deadline:
+                time.sleep(3)
+            completions = sched.get_job_completion_times()
+            assert len(completions) == len(jobs), (
+                f"only {len(completions)}/{len(jobs)} jobs completed"
+            )
+            steps = sched.get_completed_steps()
+            for j, (jid, done) in zip(jobs, sorted(steps.items())):
+                assert done >= 1, f"job {jid} made no progress"
+        finally:
+            sched.shutdown()

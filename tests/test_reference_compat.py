@@ -210,3 +210,82 @@ class TestReferenceParity:
         unfair = sum(1 for r in ftf if r > 1.05)
         ref_unfair = sum(1 for r in ref_rhos if r > 1.05)
         assert unfair == ref_unfair
+
+
+class TestProtoSchemaParity:
+    """Our .proto files (shockwave_amd/rpc/protos/) must carry the
+    reference's field names and numbers for every shared message, and the
+    msgpack transport must key its maps by those field names.  protoc is
+    not in this image; when it exists, `protoc --python_out` over our
+    protos yields wire-compatible stubs."""
+
+    OURS = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "shockwave_amd", "rpc", "protos",
+    )
+    REF_DIR = os.path.join(REFERENCE, "runtime", "protobuf")
+
+    @staticmethod
+    def _parse(path):
+        import re
+
+        msgs = {}
+        cur = None
+        for line in open(path):
+            line = line.split("//")[0].strip()
+            m = re.match(r"message (\w+) \{", line)
+            if m:
+                cur = m.group(1)
+                msgs[cur] = {}
+                continue
+            if line.startswith("}"):
+                cur = None
+                continue
+            f = re.match(
+                r"(?:repeated )?[\w.]+ (\w+) = (\d+);", line
+            )
+            if cur and f:
+                msgs[cur][f.group(1)] = int(f.group(2))
+        return msgs
+
+    def test_message_fields_match_reference(self):
+        if not os.path.isdir(self.REF_DIR):
+            pytest.skip("reference tree not present")
+        ours, ref = {}, {}
+        for d, out in ((self.OURS, ours), (self.REF_DIR, ref)):
+            for name in os.listdir(d):
+                if name.endswith(".proto"):
+                    out.update(self._parse(os.path.join(d, name)))
+        shared = set(ours) & set(ref)
+        # every reference message must exist in ours
+        assert set(ref) - {"Empty", "JobState"} <= set(ours), (
+            set(ref) - set(ours)
+        )
+        for msg in shared:
+            for fname, fnum in ref[msg].items():
+                assert ours[msg].get(fname) == fnum, (
+                    f"{msg}.{fname}: ours {ours[msg].get(fname)} "
+                    f"!= ref {fnum}"
+                )
+
+    def test_transport_uses_proto_field_names(self):
+        """The msgpack maps sent by the clients use exactly the proto
+        field names (schema-level wire compatibility)."""
+        ours = {}
+        for name in os.listdir(self.OURS):
+            if name.endswith(".proto"):
+                ours.update(self._parse(os.path.join(self.OURS, name)))
+        import inspect
+
+        from shockwave_amd.rpc import services
+
+        src = inspect.getsource(services)
+        for msg, fields in ours.items():
+            if msg in ("Empty", "JobState", "JobDescription", "Heartbeat"):
+                continue
+            for fname in fields:
+                if fname in ("job_state",):
+                    continue
+                assert f'"{fname}"' in src, (
+                    f"field {msg}.{fname} not used by services.py"
+                )
