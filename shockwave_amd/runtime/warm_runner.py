@@ -101,6 +101,11 @@ def _child_main():
         rc, error = _run_one(req, base_env, base_cwd, state)
         proto.write(json.dumps({"rc": rc, "error": error}) + "\n")
         proto.flush()
+        if rc == 143:
+            # SIGTERM aborted the job mid-flight (possibly mid-import):
+            # the interpreter may hold partially-initialized modules, so
+            # retire this runner; the pool spawns a fresh one
+            break
     os._exit(0)
 
 

@@ -53,12 +53,21 @@ def init_device_and_distributed(args) -> torch.device:
     else:
         device = torch.device("cpu")
     if args.world_size > 1 and not dist.is_initialized():
+        import datetime
+
         os.environ.setdefault("MASTER_ADDR", args.master_addr or "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", str(args.master_port or 29500))
+        # bounded rendezvous: a peer rank whose GPU slot is still held by
+        # an over-running lease may never arrive this round; failing fast
+        # turns the round into a zero-step micro-task failure (retried by
+        # the scheduler) instead of wedging the slot for the backend's
+        # default timeout
+        timeout_s = float(os.environ.get("SWQ_RENDEZVOUS_TIMEOUT", "60"))
         dist.init_process_group(
             backend="nccl" if use_cuda else "gloo",
             world_size=args.world_size,
             rank=args.rank,
+            timeout=datetime.timedelta(seconds=timeout_s),
         )
     return device
 

@@ -488,9 +488,9 @@ class TestEightWorkerDressRehearsal:
              "recommendation", "-n"),
         ]
         jobs = []
-        for i in range(16):
+        for i in range(12):
             jt, cmd, wd, steps_arg = specs[i % len(specs)]
-            sf = rng.choices([1, 2, 4], weights=[0.6, 0.3, 0.1])[0]
+            sf = rng.choices([1, 2, 4], weights=[0.7, 0.25, 0.05])[0]
             mode = rng.choice(["static", "static", "gns", "accordion"])
             if "Recommendation" in jt:
                 sf = 1  # not distributed in the reference
@@ -530,7 +530,7 @@ class TestEightWorkerDressRehearsal:
                 ))
             for j in jobs:
                 sched.add_job(j)
-            deadline = time.time() + 540
+            deadline = time.time() + 720
             while not sched.is_done() and time.time() < deadline:
                 time.sleep(3)
             completions = sched.get_job_completion_times()
