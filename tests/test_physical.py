@@ -490,14 +490,16 @@ class TestEightWorkerDressRehearsal:
         jobs = []
         for i in range(12):
             jt, cmd, wd, steps_arg = specs[i % len(specs)]
-            sf = rng.choices([1, 2, 4], weights=[0.7, 0.25, 0.05])[0]
+            sf = rng.choices([1, 2], weights=[0.7, 0.3])[0]
+            # sf=4 = 50% of this 8-slot cluster; the reference caps
+            # job size at 25% of its 32-GPU cluster — mirrored here
             mode = rng.choice(["static", "static", "gns", "accordion"])
             if "Recommendation" in jt:
                 sf = 1  # not distributed in the reference
             jobs.append(Job(
                 job_id=None, job_type=jt, command=cmd,
                 working_directory=wd, num_steps_arg=steps_arg,
-                total_steps=rng.randint(6, 12) * sf,
+                total_steps=rng.randint(4, 8) * sf,
                 duration=600, scale_factor=sf, mode=mode,
             ))
         profiles = [trace_mod.build_job_profile(j, throughputs) for j in jobs]
@@ -508,11 +510,11 @@ class TestEightWorkerDressRehearsal:
             port=port,
             expected_num_workers=8,
             throughputs=throughputs,
-            time_per_iteration=30,
+            time_per_iteration=40,
             profiles=profiles,
             worker_type="mi355x",
-            heartbeat_timeout_s=120.0,
-            completion_buffer_s=90.0,
+            heartbeat_timeout_s=180.0,
+            completion_buffer_s=120.0,
         )
         workers = []
         try:
