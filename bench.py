@@ -140,7 +140,28 @@ def main():
         if channels_last:
             static_x = static_x.to(memory_format=torch.channels_last)
         static_y = torch.zeros(bs, dtype=torch.long, device=device)
+        if distributed:
+            # quiesce all ranks before the capture attempt: in-flight eager
+            # work polled by the RCCL watchdog during another rank's stream
+            # capture is the classic capture-invalidation trigger
+            import torch.distributed as dist
+
+            dist.barrier()
+            torch.cuda.synchronize()
         graphed = try_graph_step(compute_step, [static_x, static_y])
+        if distributed:
+            # capture success is timing-dependent per rank, but every rank
+            # must run the SAME per-step collective schedule (full graph
+            # replays captured RCCL ops; the fallbacks issue them eagerly).
+            # Agree on the lowest common mode so modes never mix.
+            import torch.distributed as dist
+
+            ok = torch.tensor(
+                [1.0 if graphed is not None else 0.0], device=device
+            )
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 1.0:
+                graphed = None
         if graphed is not None:
             graph_mode = "full"
         elif distributed:
