@@ -148,20 +148,11 @@ def main():
 
             dist.barrier()
             torch.cuda.synchronize()
-        graphed = try_graph_step(compute_step, [static_x, static_y])
-        if distributed:
-            # capture success is timing-dependent per rank, but every rank
-            # must run the SAME per-step collective schedule (full graph
-            # replays captured RCCL ops; the fallbacks issue them eagerly).
-            # Agree on the lowest common mode so modes never mix.
-            import torch.distributed as dist
+        from shockwave_amd.parallel.graphs import agree_capture
 
-            ok = torch.tensor(
-                [1.0 if graphed is not None else 0.0], device=device
-            )
-            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
-            if ok.item() < 1.0:
-                graphed = None
+        graphed = agree_capture(
+            try_graph_step(compute_step, [static_x, static_y])
+        )
         if graphed is not None:
             graph_mode = "full"
         elif distributed:

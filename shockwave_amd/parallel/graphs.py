@@ -75,6 +75,40 @@ class GraphedTrainStep:
         return self.replay()
 
 
+def agree_capture(graphed, group=None):
+    """Collective agreement on hipGraph use at world>1.
+
+    Capture success is timing-dependent per rank (the RCCL watchdog
+    polling another rank's in-flight work can invalidate a capture), but
+    every rank must run the SAME per-step collective schedule: a full
+    graph replays its captured RCCL ops while the fallbacks issue them
+    eagerly, and mixing the two across ranks deadlocks the communicator.
+    MIN-reduce a success flag and keep the graph only if every rank
+    captured; the discarded graph is simply never replayed (no real
+    collective ran during its capture), so discarding is safe.
+
+    Returns ``graphed`` when all ranks captured, else ``None``.
+    """
+    import torch.distributed as dist
+
+    if not dist.is_initialized() or dist.get_world_size(group) == 1:
+        return graphed
+    backend = dist.get_backend(group)
+    device = torch.device("cuda") if backend == "nccl" else torch.device("cpu")
+    ok = torch.tensor([1.0 if graphed is not None else 0.0], device=device)
+    dist.all_reduce(ok, op=dist.ReduceOp.MIN, group=group)
+    if ok.item() < 1.0:
+        if graphed is not None:
+            import logging
+
+            logging.getLogger("shockwave_amd.graphs").warning(
+                "hipGraph captured here but failed on a peer rank; "
+                "discarding for a uniform collective schedule"
+            )
+        return None
+    return graphed
+
+
 def try_graph_step(step_fn, static_inputs, warmup_iters=3):
     """Capture if possible; return None when capture is unsupported for
     this step (e.g. RCCL build without graph support) so callers keep the

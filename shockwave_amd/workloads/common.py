@@ -63,8 +63,14 @@ def init_device_and_distributed(args) -> torch.device:
         # the scheduler) instead of wedging the slot for the backend's
         # default timeout
         timeout_s = float(os.environ.get("SWQ_RENDEZVOUS_TIMEOUT", "60"))
+        # SWQ_DIST_BACKEND forces the backend: RCCL refuses two ranks on
+        # one device (documented in profiles/MULTIGPU_PROBE.md), so
+        # time-sliced measurement runs world>1 on a single GPU over gloo
+        backend = os.environ.get(
+            "SWQ_DIST_BACKEND", "nccl" if use_cuda else "gloo"
+        )
         dist.init_process_group(
-            backend="nccl" if use_cuda else "gloo",
+            backend=backend,
             world_size=args.world_size,
             rank=args.rank,
             timeout=datetime.timedelta(seconds=timeout_s),

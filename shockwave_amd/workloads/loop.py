@@ -255,6 +255,10 @@ def run(spec: WorkloadSpec, args, mode: Optional[str] = None, client=None,
             warmup = 3 if gns is None else max(3, gns.window + 1)
             graphed = try_graph_step(lambda: graph_body(), [],
                                      warmup_iters=warmup)
+            if args.world_size > 1:
+                from ..parallel.graphs import agree_capture
+
+                graphed = agree_capture(graphed)
             with torch.no_grad():
                 model.load_state_dict(pre)
             sess.graphed, sess.static_batch = graphed, static_batch

@@ -200,3 +200,63 @@ class TestBucketAlignment:
         (ref(xa).sum() + ref(xb).sum()).backward()
         for p, r in zip(model.parameters(), ref.parameters()):
             torch.testing.assert_close(p.grad, r.grad)
+
+
+def _agree_worker(rank, world_size, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world_size)
+        from shockwave_amd.parallel.graphs import agree_capture
+
+        sentinel = object()
+        # round 1: only rank 0 "captured" -> everyone must discard
+        mine = sentinel if rank == 0 else None
+        r1 = agree_capture(mine)
+        # round 2: every rank captured -> everyone keeps its graph
+        r2 = agree_capture(sentinel)
+        q.put((rank, r1 is None, r2 is sentinel))
+        dist.destroy_process_group()
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put((rank, "ERROR", traceback.format_exc()))
+
+
+class TestAgreeCapture:
+    """All ranks must converge on one hipGraph mode (bench.py / loop.py
+    wire this in at world>1 so captured RCCL schedules never mix with
+    eager fallbacks across ranks)."""
+
+    def test_world2_agreement(self):
+        import socket
+
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        procs = [
+            ctx.Process(target=_agree_worker, args=(r, 2, port, q))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        results = [q.get() for _ in range(2)]
+        for p in procs:
+            p.join(timeout=120)
+            assert p.exitcode == 0
+        for r in results:
+            assert r[1] != "ERROR", r[2]
+            assert r[1] is True   # disagreement -> discarded everywhere
+            assert r[2] is True   # unanimity -> kept
+
+    def test_world1_passthrough(self):
+        from shockwave_amd.parallel.graphs import agree_capture
+
+        s = object()
+        assert agree_capture(s) is s  # no process group: no-op
