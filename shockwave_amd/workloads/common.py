@@ -55,8 +55,17 @@ def init_device_and_distributed(args) -> torch.device:
     if args.world_size > 1 and not dist.is_initialized():
         import datetime
 
-        os.environ.setdefault("MASTER_ADDR", args.master_addr or "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", str(args.master_port or 29500))
+        # explicit dispatcher args take precedence over a stale env (a
+        # warm runner re-inits per lease; reusing the previous lease's
+        # port can connect to its dying TCPStore and hang rendezvous)
+        if args.master_addr:
+            os.environ["MASTER_ADDR"] = args.master_addr
+        else:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        if args.master_port:
+            os.environ["MASTER_PORT"] = str(args.master_port)
+        else:
+            os.environ.setdefault("MASTER_PORT", "29500")
         # bounded rendezvous: a peer rank whose GPU slot is still held by
         # an over-running lease may never arrive this round; failing fast
         # turns the round into a zero-step micro-task failure (retried by
