@@ -75,9 +75,19 @@ def main():
     ap.add_argument("--oracle", default="traces/mi355x_throughputs.json")
     ap.add_argument("--out_oracle", required=True)
     ap.add_argument("--out_startup", required=True)
+    ap.add_argument("--trace", default=None,
+                    help="trace file: recover job_id -> job_type by line "
+                         "order when the pickle predates the job_types "
+                         "field (run_physical.py)")
     args = ap.parse_args()
 
     results = pickle.load(open(args.physical, "rb"))
+    if not results.get("job_types") and args.trace:
+        results["job_types"] = {
+            str(i): line.split("\t")[0]
+            for i, line in enumerate(open(args.trace))
+            if line.strip()
+        }
     hot, startup = calibrate(results)
     print("calibrated hot rates:", json.dumps(hot, indent=1))
     print("first-dispatch startup:", json.dumps(startup, indent=1))

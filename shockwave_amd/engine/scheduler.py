@@ -813,6 +813,14 @@ class RoundScheduler:
                 self._worker_time_so_far.get(wt, 0.0) - d
             )
         try:
+            # our physical mechanism recomputes the allocation at EVERY
+            # mid-round (physical.py _mid_round: _update_priorities +
+            # _compute_allocation before _schedule_jobs_on_workers); the
+            # event simulator's default throttle (min 1000 s between
+            # resets, reference parity) starves jobs that arrive between
+            # resets out of the allocation.  Mirror the physical cadence.
+            self._allocation = self._compute_allocation()
+            self._need_to_update_allocation = False
             return self._schedule_jobs_on_workers()
         finally:
             if self._last_reset_time == pre_reset:
@@ -1553,6 +1561,14 @@ class RoundScheduler:
                 # recompute so the clock can advance
                 self._need_to_update_allocation = True
                 self._last_reset_time = -self._minimum_time_between_allocation_resets
+                # the failed attempt recorded an empty round entry and
+                # counted every job as queued; the recompute below records
+                # THIS round's real schedule — undo the empty record
+                if self._per_round_schedule and not self._per_round_schedule[-1]:
+                    self._per_round_schedule.pop()
+                    self._num_jobs_in_curr_round.pop()
+                    for jid_q in self._jobs:
+                        self._num_queued_rounds[jid_q] -= 1
                 scheduled_jobs = self._schedule_jobs_on_workers()
             if not scheduled_jobs and self._jobs and not queued_jobs:
                 # jobs larger than the cluster can never run: fail them
