@@ -367,12 +367,25 @@ class _RolloutLoader:
 def rl_main(argv=None, mode=None, client=None, max_steps_override=None):
     p = argparse.ArgumentParser()
     p.add_argument("--env", type=str, default="PongDeterministic-v4")
-    p.add_argument("--workers", type=int, default=4)
+    p.add_argument("--workers", type=int, default=4,
+                   help="Hogwild actor processes; 0 = single-process "
+                        "rollout trainer")
+    p.add_argument("--rollout", type=int, default=20,
+                   help="env frames per actor update (A3C num-steps)")
     p.add_argument("--amsgrad", type=str, default="True")
     p.add_argument("--lr", type=float, default=1e-4)
     common.add_scheduler_args(p, "--max-steps")
     args = p.parse_args(argv)
     args.batch_size = 4
+
+    if args.workers > 0:
+        # the reference's default path: lock-free shared-memory actors
+        # (rl/main.py:224 torch.multiprocessing + SharedAdam)
+        from .hogwild import hogwild_train
+
+        return hogwild_train(
+            args, client=client, max_steps_override=max_steps_override
+        )
 
     def step(model, batch, device, state):
         m = model.module if hasattr(model, "module") else model

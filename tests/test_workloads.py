@@ -44,9 +44,31 @@ class TestFamilies:
         steps = families.cyclegan_main(["--n_steps", "1"])
         assert steps == 1
 
-    def test_rl(self):
-        steps = families.rl_main(["--max-steps", "3"])
+    def test_rl_single_process(self):
+        steps = families.rl_main(["--max-steps", "3", "--workers", "0"])
         assert steps == 3
+
+    def test_rl_hogwild(self):
+        # the reference's default rl path: torch.multiprocessing actors
+        # around a share_memory() model + SharedAdam (rl/main.py:224)
+        steps = families.rl_main(
+            ["--max-steps", "3", "--workers", "2", "--rollout", "5"]
+        )
+        assert steps == 3
+
+    def test_rl_hogwild_checkpoint_resume(self, tmp_path):
+        args = ["--workers", "2", "--rollout", "5",
+                "--checkpoint_dir", str(tmp_path),
+                "--enable_gavel_iterator"]
+        from shockwave_amd.runtime.lease_iterator import NullLeaseClient
+
+        s1 = families.rl_main(args + ["--max-steps", "2"],
+                              client=NullLeaseClient())
+        assert s1 == 2
+        # second lease resumes from the checkpointed cumulative count
+        s2 = families.rl_main(args + ["--max-steps", "5"],
+                              client=NullLeaseClient())
+        assert s2 == 5  # 2 restored + 3 more
 
 
 class TestShims:
