@@ -183,13 +183,55 @@ def print_report(report, results, oracle, sim=None):
             print(f"  job {i}: sim {s:.1f}  phys {p:.1f}  gap {gap:+.1f}%")
 
 
+def print_sim_report(sim):
+    """Standalone decoder for a simulation results pickle (the
+    reference's parse_simulation_trace.py role for SIM output): per-job
+    JCT/FTF and scheduled-round counts from the per-round schedule."""
+    print("=" * 78)
+    print(
+        f"SIM policy={sim.get('policy')} "
+        f"makespan={sim.get('makespan_s', 0):.1f}s "
+        f"avg_jct={sim.get('avg_jct_s', 0):.1f}s "
+        f"worst_rho={sim.get('worst_ftf_rho')}"
+    )
+    print("=" * 78)
+    prs = sim.get("per_round_schedule") or []
+    sched_rounds = defaultdict(list)
+    for i, rd in enumerate(prs):
+        for jid in rd:
+            sched_rounds[str(jid)].append(i)
+    jcts = sim.get("jct_list", [])
+    rhos = sim.get("ftf_rho_list", [])
+    for i, jct in enumerate(jcts):
+        rounds = sched_rounds.get(str(i), [])
+        rho = rhos[i] if i < len(rhos) else None
+        rtxt = (
+            f"rounds {rounds[0]}-{rounds[-1]} ({len(rounds)} scheduled)"
+            if rounds else "never scheduled"
+        )
+        print(f"job {i}: jct {jct:8.1f}s  rho {rho}  {rtxt}")
+    if prs:
+        width = max((len(str(j)) for rd in prs for j in rd), default=1)
+        print("-" * 78)
+        print("round -> jobs:")
+        for i, rd in enumerate(prs):
+            jobs = " ".join(str(j).rjust(width) for j in sorted(rd))
+            print(f"  r{i:03d}: {jobs if jobs else '(idle)'}")
+
+
 def main():
     ap = argparse.ArgumentParser(description=__doc__)
-    ap.add_argument("--physical", required=True)
+    ap.add_argument("--physical", default=None)
     ap.add_argument("--simulation", default=None)
     ap.add_argument("--oracle", default=None)
     ap.add_argument("--json_out", default=None)
     args = ap.parse_args()
+
+    if args.physical is None:
+        if args.simulation is None:
+            ap.error("need --physical and/or --simulation")
+        print_sim_report(pickle.load(open(args.simulation, "rb")))
+        return
 
     results = pickle.load(open(args.physical, "rb"))
     oracle = json.load(open(args.oracle)) if args.oracle else None
