@@ -38,6 +38,7 @@ def run_simulation(
     fixed_rounds=False,
     startup_table=None,
     ideal=False,
+    world_throughputs_file=None,
 ):
     import logging
 
@@ -89,6 +90,8 @@ def run_simulation(
         midround_staleness=midround_staleness,
         fixed_rounds=fixed_rounds,
         startup_table=startup_table,
+        world_throughputs=(read_throughputs(world_throughputs_file)
+                           if world_throughputs_file else None),
     )
 
     start = time.time()
@@ -164,6 +167,11 @@ def main():
                    help="model the physical planner's mid-round decision "
                         "point (half of the current round's service "
                         "unobserved) for non-shockwave policies")
+    p.add_argument("--world_throughputs_file", default=None,
+                   help="fidelity: rates the simulated WORLD runs at, "
+                        "while --throughputs_file stays the rates the "
+                        "policy believes (the physical scheduler's own "
+                        "oracle)")
     p.add_argument("--startup_table", default=None,
                    help="JSON of per-job-type first-dispatch startup "
                         "seconds (scripts/calibrate_sim.py)")
@@ -215,6 +223,7 @@ def main():
         startup_table=(json.load(open(args.startup_table))
                        if args.startup_table else None),
         ideal=args.ideal,
+        world_throughputs_file=args.world_throughputs_file,
     )
     print(
         json.dumps(

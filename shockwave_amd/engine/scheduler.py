@@ -78,10 +78,12 @@ class RoundScheduler:
         warm_preemption_overhead_s: Optional[float] = None,
         fixed_rounds: bool = False,
         startup_table: Optional[Dict[str, float]] = None,
+        world_throughputs: Optional[Dict] = None,
     ):
         self._policy = policy
         self._simulate = simulate
         self._oracle_throughputs = throughputs
+        self._world_throughputs = world_throughputs
         self._midround_staleness = midround_staleness
         self._time_per_iteration = time_per_iteration
         self._minimum_time_between_allocation_resets = (
@@ -920,6 +922,26 @@ class RoundScheduler:
     # Step accounting (reference :1425-1516)
     # ------------------------------------------------------------------
 
+    def _world_rate(self, job_id, worker_type):
+        """Rate the simulated WORLD runs at, when it differs from the
+        rate the POLICY believes.  Physically the two are distinct: the
+        scheduler plans with its oracle while jobs progress at the box's
+        measured rates — a fidelity sim must keep both tables or it
+        either mis-clocks the world (belief-only) or leaks ground truth
+        into the policy's decisions (world-only).  None = no split
+        (reference-parity behavior: one table for both)."""
+        if self._world_throughputs is None or job_id.is_pair():
+            return None
+        job = self._jobs.get(job_id)
+        if job is None:
+            return None
+        entry = self._world_throughputs.get(worker_type, {}).get(
+            (job.job_type, job.scale_factor)
+        )
+        if entry:
+            return entry.get("null")
+        return None
+
     def _get_num_steps(self, job_id, worker_type, single_job_id=None,
                        startup_s=0.0):
         effective_time = max(0.0, self._time_per_iteration - startup_s)
@@ -930,7 +952,10 @@ class RoundScheduler:
                 self._throughputs[job_id][worker_type][index] * effective_time
             )
         else:
-            tput = self._throughputs[job_id][worker_type]
+            tput = (
+                self._world_rate(job_id, worker_type)
+                or self._throughputs[job_id][worker_type]
+            )
             if job_id.is_pair():
                 index = 0 if job_id.singletons()[0] == single_job_id else 1
                 tput = tput[index]
@@ -954,7 +979,10 @@ class RoundScheduler:
                 job_id, worker_type, single, startup_s=startup_s
             )
             all_num_steps.append(num_steps)
-            tput = self._throughputs[job_id][worker_type]
+            tput = (
+                self._world_rate(job_id, worker_type)
+                or self._throughputs[job_id][worker_type]
+            )
             if job_id.is_pair():
                 index = 0 if job_id.singletons()[0] == single else 1
                 tput = tput[index]
@@ -985,8 +1013,12 @@ class RoundScheduler:
                 tput,
                 bs,
             )
-        if not self._simulate:
-            # EMA between old value and new measurement (reference :596-601)
+        if not self._simulate or self._world_throughputs is not None:
+            # EMA between old value and new measurement (reference
+            # :596-601).  In a belief/world-split fidelity sim the same
+            # online calibration runs: the policy's belief converges to
+            # the observed (world) rate exactly as the physical
+            # scheduler's does
             for i, single in enumerate(job_id.singletons()):
                 if all_execution_times[i] <= 0:
                     continue
