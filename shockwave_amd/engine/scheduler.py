@@ -959,6 +959,21 @@ class RoundScheduler:
             if job_id.is_pair():
                 index = 0 if job_id.singletons()[0] == single_job_id else 1
                 tput = tput[index]
+            if self._fixed_rounds and not job_id.is_pair():
+                # fidelity mode: the physical iterator self-completes when
+                # projected run time crosses the deadline, mid-lease
+                # (lease_iterator.py:292-296) — clip this micro-task at
+                # the crossing instead of running the full round (the
+                # +0.5 s lands run_time just past the `>` deadline check
+                # in the done callback, as the real abort does)
+                job = self._jobs.get(job_id)
+                if job is not None and getattr(job, "duration", 0):
+                    run_so_far = sum(
+                        self._cumulative_run_time.get(job_id, {}).values()
+                    ) / max(1, job.scale_factor)
+                    remaining_dl = int(job.duration * 1.5) - run_so_far
+                    if remaining_dl < effective_time:
+                        effective_time = max(0.0, remaining_dl) + 0.5
             num_steps = int(tput * effective_time)
         target = single_job_id if single_job_id is not None else job_id
         return min(num_steps, self._get_remaining_steps(target))
