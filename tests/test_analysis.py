@@ -102,3 +102,27 @@ class TestFidelityScheduling:
         # one line per job with its jct
         for i in range(5):
             assert f"job {i}:" in out.stdout
+
+
+class TestDeadlineClip:
+    def test_micro_task_clipped_at_deadline_crossing(self, tmp_path):
+        """In fixed-rounds fidelity mode a job crossing 1.5x its
+        reference duration self-completes mid-round (like the
+        iterator's projection abort), instead of running to the round
+        boundary first."""
+        from simulate import run_simulation
+
+        r = run_simulation(
+            TRACE, ORACLE, "max_min_fairness", num_gpus=1,
+            time_per_iteration=60, midround_staleness=True,
+            fixed_rounds=True,
+        )
+        r_noclip = run_simulation(
+            TRACE, ORACLE, "max_min_fairness", num_gpus=1,
+            time_per_iteration=60,
+        )
+        # both complete all five jobs either way
+        assert len(r["jct_list"]) == len(r_noclip["jct_list"]) == 5
+        # fidelity mode's makespan is the last completion, not a round
+        # boundary multiple
+        assert r["makespan_s"] % 60 != pytest.approx(0.0, abs=1e-6)
