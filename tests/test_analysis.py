@@ -126,3 +126,48 @@ class TestDeadlineClip:
         # fidelity mode's makespan is the last completion, not a round
         # boundary multiple
         assert r["makespan_s"] % 60 != pytest.approx(0.0, abs=1e-6)
+
+
+class TestWorldSplit:
+    def test_world_rates_clock_the_sim_not_the_policy(self, tmp_path):
+        """Belief/world throughput split: with a world oracle at half
+        the believed rates, jobs take about twice as long — the sim's
+        clock follows the WORLD table while the policy still plans on
+        its beliefs (the physical scheduler's situation)."""
+        import json
+
+        from simulate import run_simulation
+
+        belief = run_simulation(
+            TRACE, ORACLE, "max_min_fairness", num_gpus=1,
+            time_per_iteration=60, midround_staleness=True,
+            fixed_rounds=True,
+        )
+        slow = json.load(open(ORACLE))
+        for table in slow.values():
+            for entry in table.values():
+                if isinstance(entry, dict) and "null" in entry:
+                    entry["null"] = (
+                        entry["null"] / 2.0
+                        if isinstance(entry["null"], (int, float))
+                        else entry["null"]
+                    )
+        world_path = tmp_path / "slow_world.json"
+        with open(world_path, "w") as f:
+            json.dump(slow, f)
+        split = run_simulation(
+            TRACE, ORACLE, "max_min_fairness", num_gpus=1,
+            time_per_iteration=60, midround_staleness=True,
+            fixed_rounds=True, world_throughputs_file=str(world_path),
+        )
+        # jobs need more rounds at half speed...
+        assert split["makespan_s"] > belief["makespan_s"]
+        growth = max(
+            s_ / b_ for s_, b_ in zip(split["jct_list"], belief["jct_list"])
+        )
+        assert growth > 1.10
+        # ...but the slowdown is CAPPED by the (faithful) deadline clip:
+        # at half the believed rate every job crosses 1.5x its believed
+        # duration and self-completes, exactly as the iterator would
+        assert split["makespan_s"] < 2.0 * belief["makespan_s"]
+        assert len(split["jct_list"]) == len(belief["jct_list"]) == 5

@@ -227,3 +227,16 @@ class TestGNSGraphCapture:
             graph.replay()
         torch.cuda.synchronize()
         torch.testing.assert_close(est._avg, ref_avg, rtol=1e-4, atol=1e-6)
+
+
+class TestHogwildOnGPU:
+    def test_hogwild_actors_train_on_device(self):
+        """Round-2 Hogwild A3C: multiprocessing actors each running
+        forward/backward on the MI355X against a shared CPU model
+        (workloads/hogwild.py; reference rl/main.py:224)."""
+        from shockwave_amd.workloads import families
+
+        steps = families.rl_main(
+            ["--max-steps", "4", "--workers", "2", "--rollout", "5"]
+        )
+        assert steps == 4
