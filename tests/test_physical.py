@@ -41,7 +41,13 @@ def tiny_job(steps=6, bs=16):
 
 @pytest.mark.slow
 class TestPhysicalEndToEnd:
-    def test_single_job_completes_through_rpc(self, tmp_path, throughputs):
+    @pytest.mark.parametrize("codec", ["msgpack", "proto"])
+    def test_single_job_completes_through_rpc(self, tmp_path, throughputs,
+                                              codec, monkeypatch):
+        # "proto" runs the whole control plane on TRUE protobuf wire
+        # bodies from the runtime-built stubs (rpc/pb.py) — the
+        # reference's protoc wire format, kept green in every CPU run
+        monkeypatch.setenv("SWQ_RPC_CODEC", codec)
         from shockwave_amd.engine.physical import PhysicalScheduler
         from shockwave_amd.policies import get_policy
         from shockwave_amd.runtime.worker import Worker
